@@ -1,0 +1,166 @@
+/* vega_gpu.h — the drop-in C ABI of the MI355X-native shuffle/sort/aggregate
+ * engine (libvega_gpu.so).
+ *
+ * DROP-IN BOUNDARY: the reference's GPU-replaceable seam is
+ *   - map side:  ShuffleDependencyTrait::do_shuffle_task(rdd, partition) -> uri
+ *                (/root/reference/src/dependency.rs:92-97, hot loop :164-229)
+ *   - reduce side: Rdd::compute(split) -> iterator
+ *                (/root/reference/src/rdd/rdd.rs:179; shuffled_rdd.rs:149-170)
+ * A Rust host keeps the Rdd/PairRdd trait surface and calls these entry
+ * points over plain FFI (see INTEGRATION.md for the exact `extern "C"` block
+ * a vega maintainer would add). No torch types, no C++ types: pointers,
+ * sizes, int error codes. Handles are opaque.
+ *
+ * Two API levels:
+ *  1. RDD-handle API — mirrors Context::parallelize/make_rdd
+ *     (context.rs:406-442) + PairRdd ops (pair_rdd.rs:20-171) for a
+ *     single-process host. One context drives ONE GPU (vega local mode is
+ *     one process; multi-GPU runs one process per GPU, rank model below).
+ *  2. Device-pointer API — for a rank-per-GPU launcher (torch.distributed /
+ *     RCCL over xGMI): caller owns device buffers and the stream; these are
+ *     the raw kernel entries (map-side radix partition replacing
+ *     dependency.rs:191-210, sort+segmented-reduce replacing
+ *     shuffled_rdd.rs:154-164's HashMap merge).
+ *
+ * All functions return 0 on success, negative VEGA_ERR_* otherwise.
+ */
+#ifndef VEGA_GPU_H
+#define VEGA_GPU_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define VEGA_OK               0
+#define VEGA_ERR_INVALID     -1
+#define VEGA_ERR_NOMEM       -2
+#define VEGA_ERR_HIP         -3
+#define VEGA_ERR_UNSUPPORTED -4
+#define VEGA_ERR_CAP         -5   /* output capacity too small */
+
+/* Aggregator op-enum: the fixed-function replacements for the reference's
+ * boxed closures (aggregator.rs:8-16; reduce_by_key closures
+ * pair_rdd.rs:74-78; group default aggregator.rs:33-53). */
+typedef enum {
+    VEGA_OP_SUM_I64 = 0,   /* reduce_by_key(+) on i64 (wrapping, bit-exact) */
+    VEGA_OP_COUNT   = 1,   /* group_by_key -> per-key count */
+    VEGA_OP_SUM_F64 = 2,   /* reduce_by_key(+) on f64 (1e-6 rel tolerance) */
+    VEGA_OP_MIN_I64 = 3,
+    VEGA_OP_MAX_I64 = 4,
+} vega_op_t;
+
+typedef struct vega_ctx vega_ctx_t;
+typedef uint64_t vega_rdd_t;      /* opaque RDD handle, 0 = invalid */
+
+/* ---------------- context ---------------- */
+/* ngpus: number of GPUs this process drives. The rank-per-GPU model uses 1
+ * (device = current HIP device); values > 1 are reserved. */
+int vega_gpu_init(int ngpus, vega_ctx_t **out);
+int vega_gpu_shutdown(vega_ctx_t *ctx);
+int vega_gpu_synchronize(vega_ctx_t *ctx);
+const char *vega_gpu_last_error(vega_ctx_t *ctx);
+
+/* ---------------- RDD construction ---------------- */
+/* make_rdd: host (k,v) arrays copied to device; nparts = logical partition
+ * count with ParallelCollection::slice chunking
+ * (parallel_collection_rdd.rs:116-145): partition p = rows [pn/P,(p+1)n/P). */
+int vega_gpu_make_rdd(vega_ctx_t *ctx, const int64_t *keys, const int64_t *vals,
+                      uint64_t n, uint32_t nparts, vega_rdd_t *out);
+int vega_gpu_make_rdd_f64(vega_ctx_t *ctx, const int64_t *keys, const double *vals,
+                          uint64_t n, uint32_t nparts, vega_rdd_t *out);
+/* device-side deterministic generation (bit-identical to
+ * vega_gen_uniform_pairs_i64 in datagen.c); start = global row offset so each
+ * rank generates its own shard of one global stream. */
+int vega_gpu_gen_rdd_uniform(vega_ctx_t *ctx, uint64_t n, uint64_t seed,
+                             int key_bits, uint64_t start, uint32_t nparts,
+                             vega_rdd_t *out);
+
+/* ---------------- PairRdd ops (pair_rdd.rs names) ---------------- */
+/* reduce_by_key (pair_rdd.rs:54-80) with op as the aggregator */
+int vega_gpu_reduce_by_key(vega_ctx_t *ctx, vega_rdd_t rdd, vega_op_t op,
+                           uint32_t nparts, vega_rdd_t *out);
+/* group_by_key -> (key, group size) (cfg C2; full group materialization is
+ * host-side via collect of the sorted pairs) */
+int vega_gpu_group_count(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
+                         vega_rdd_t *out);
+/* sort_by_key ascending, signed i64 order, stable (absent from the
+ * reference — SURVEY.md §8a a8; Spark semantics) */
+int vega_gpu_sort_by_key(vega_ctx_t *ctx, vega_rdd_t rdd, vega_rdd_t *out);
+/* inner join via sorted runs (co_grouped_rdd.rs:206-249 + pair_rdd.rs:104-121) */
+int vega_gpu_join(vega_ctx_t *ctx, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
+                  vega_rdd_t *out);
+/* distinct (rdd.rs:501-531): keys of rdd deduplicated */
+int vega_gpu_distinct(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
+                      vega_rdd_t *out);
+
+/* ---------------- actions ---------------- */
+int vega_gpu_count(vega_ctx_t *ctx, vega_rdd_t rdd, uint64_t *n);
+/* collect (rdd.rs:420-434): D2H of the rows. Call with keys==NULL to query n. */
+int vega_gpu_collect(vega_ctx_t *ctx, vega_rdd_t rdd, int64_t *keys, void *vals,
+                     uint64_t *n);
+int vega_gpu_free_rdd(vega_ctx_t *ctx, vega_rdd_t rdd);
+
+/* ---------------- profiling ---------------- */
+int vega_gpu_set_profiling(vega_ctx_t *ctx, int enabled);
+/* JSON {"kernel":{"ms":total_ms,"n":launches},...} since last enable */
+int vega_gpu_kernel_stats(vega_ctx_t *ctx, char *buf, size_t buflen);
+
+/* =========== device-pointer API (rank-per-GPU launcher) =========== */
+/* stream: a hipStream_t cast to void*; all calls async on that stream unless
+ * noted. Buffers are caller-owned DEVICE pointers (e.g. torch tensors). */
+
+/* workspace bytes required for n rows (sort temps + scan scratch) */
+size_t vega_dev_ws_bytes(uint64_t n);
+
+/* deterministic uniform generator kernel (rows [start, start+n) of stream
+ * `seed`, keys masked to key_bits) */
+int vega_dev_gen_uniform_i64(void *stream, int64_t *keys, int64_t *vals,
+                             uint64_t n, uint64_t seed, int key_bits,
+                             uint64_t start);
+
+/* map-side radix partition (replaces the per-row get_partition + HashMap of
+ * dependency.rs:191-210): bucket of row = splitmix64(key) % nparts; rows
+ * scattered bucket-contiguous into out_k/out_v (each n rows), per-bucket
+ * counts written to h_counts[nparts] ON THE HOST after an internal stream
+ * sync (the exchange plan needs them). nparts <= 256. */
+int vega_dev_partition_i64(void *stream, const int64_t *keys, const int64_t *vals,
+                           uint64_t n, uint32_t nparts,
+                           int64_t *out_k, int64_t *out_v,
+                           uint64_t *h_counts, void *d_ws, size_t ws_bytes);
+
+/* reduce-side sort+segmented-aggregate (replaces shuffled_rdd.rs:154-164's
+ * HashMap merge_combiners): stable LSB radix sort of (k,v) by key then one
+ * combiner per equal-key run. in_k/in_v are NOT modified. out arrays must
+ * hold n rows (worst case all-distinct). *h_nout = #distinct keys (host,
+ * after internal sync). vals/out_v are int64 for SUM_I64/COUNT/MIN/MAX,
+ * double for SUM_F64. */
+int vega_dev_sort_reduce(void *stream, const int64_t *in_k, const void *in_v,
+                         uint64_t n, int op, int64_t *out_k, void *out_v,
+                         uint64_t *h_nout, void *d_ws, size_t ws_bytes);
+
+/* stable LSB radix sort by signed-i64 key (sort_by_key). In-place semantics:
+ * result lands back in keys/vals. */
+int vega_dev_sort_pairs_i64(void *stream, int64_t *keys, int64_t *vals,
+                            uint64_t n, void *d_ws, size_t ws_bytes);
+
+/* sort-merge inner join of two KEY-SORTED sides (K4): counts pass + emit
+ * pass. *h_nout = rows emitted (<= cap). out_* sized cap. */
+int vega_dev_join_sorted(void *stream,
+                         const int64_t *ak, const int64_t *av, uint64_t na,
+                         const int64_t *bk, const int64_t *bv, uint64_t nb,
+                         int64_t *out_k, int64_t *out_va, int64_t *out_vb,
+                         uint64_t cap, uint64_t *h_nout,
+                         void *d_ws, size_t ws_bytes);
+
+/* order-independent multiset checksum of device rows (same formula as
+ * oracle_checksum_pairs_i64) — large-size parity checks without D2H */
+int vega_dev_checksum_pairs(void *stream, const int64_t *keys, const int64_t *vals,
+                            uint64_t n, uint64_t *h_sum, void *d_ws, size_t ws_bytes);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* VEGA_GPU_H */

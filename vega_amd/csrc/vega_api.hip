@@ -1,0 +1,343 @@
+/* vega_api.hip — the C-ABI (include/vega_gpu.h) over the CDNA4 kernels.
+ *
+ * RDD-handle API: the single-process drop-in for the reference's
+ * Context/PairRdd seam (context.rs:406-442, pair_rdd.rs:20-171); see
+ * INTEGRATION.md for the Rust-side binding. Device-pointer API: raw entries
+ * for the rank-per-GPU launcher (torch.distributed / RCCL over xGMI).
+ *
+ * Fails loudly: every entry returns a negative VEGA_ERR_* on any HIP error;
+ * there is no CPU fallback anywhere in this library.
+ */
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <string>
+
+#include "../../include/vega_gpu.h"
+#include "vega_internal.h"
+
+using namespace vega;
+
+struct RddImpl {
+    uint64_t n = 0;
+    int vtype = 0; /* 0 = i64 vals, 1 = f64 vals */
+    int64_t *d_k = nullptr;
+    void *d_v = nullptr;
+    uint64_t alloc_rows = 0;
+    uint32_t nparts = 1;
+    bool sorted = false;
+};
+
+struct vega_ctx {
+    int device = 0;
+    hipStream_t stream = nullptr;
+    std::map<uint64_t, RddImpl *> rdds;
+    uint64_t next_id = 1;
+    void *ws = nullptr;
+    size_t ws_bytes = 0;
+    char err[512] = {0};
+};
+
+#define CTX_TRY(ctx, x)                                                        \
+    do {                                                                       \
+        hipError_t _e = (x);                                                   \
+        if (_e != hipSuccess) {                                                \
+            snprintf((ctx)->err, sizeof (ctx)->err, "%s:%d: %s", __FILE__,     \
+                     __LINE__, hipGetErrorString(_e));                         \
+            return VEGA_ERR_HIP;                                               \
+        }                                                                      \
+    } while (0)
+
+static int ensure_ws(vega_ctx *c, uint64_t n) {
+    size_t need = ws_bytes_for(n);
+    if (c->ws_bytes >= need) return VEGA_OK;
+    if (c->ws) (void)hipFree(c->ws);
+    c->ws = nullptr;
+    c->ws_bytes = 0;
+    CTX_TRY(c, hipMalloc(&c->ws, need));
+    c->ws_bytes = need;
+    return VEGA_OK;
+}
+
+static RddImpl *get_rdd(vega_ctx *c, vega_rdd_t h) {
+    auto it = c->rdds.find(h);
+    return it == c->rdds.end() ? nullptr : it->second;
+}
+
+static int new_rdd(vega_ctx *c, uint64_t rows_alloc, int vtype, uint32_t nparts,
+                   RddImpl **out, vega_rdd_t *hout) {
+    RddImpl *r = new RddImpl();
+    r->vtype = vtype;
+    r->nparts = nparts ? nparts : 1;
+    r->alloc_rows = rows_alloc;
+    if (rows_alloc) {
+        CTX_TRY(c, hipMalloc(&r->d_k, rows_alloc * 8));
+        CTX_TRY(c, hipMalloc(&r->d_v, rows_alloc * 8));
+    }
+    uint64_t h = c->next_id++;
+    c->rdds[h] = r;
+    *out = r;
+    *hout = h;
+    return VEGA_OK;
+}
+
+extern "C" {
+
+int vega_gpu_init(int ngpus, vega_ctx_t **out) {
+    if (ngpus != 1) return VEGA_ERR_UNSUPPORTED; /* rank-per-GPU model */
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess || ndev < 1) return VEGA_ERR_HIP;
+    vega_ctx *c = new vega_ctx();
+    (void)hipGetDevice(&c->device);
+    if (hipStreamCreate(&c->stream) != hipSuccess) {
+        delete c;
+        return VEGA_ERR_HIP;
+    }
+    *out = c;
+    return VEGA_OK;
+}
+
+int vega_gpu_shutdown(vega_ctx_t *c) {
+    if (!c) return VEGA_ERR_INVALID;
+    (void)hipStreamSynchronize(c->stream);
+    for (auto &kv : c->rdds) {
+        if (kv.second->d_k) (void)hipFree(kv.second->d_k);
+        if (kv.second->d_v) (void)hipFree(kv.second->d_v);
+        delete kv.second;
+    }
+    if (c->ws) (void)hipFree(c->ws);
+    (void)hipStreamDestroy(c->stream);
+    delete c;
+    return VEGA_OK;
+}
+
+int vega_gpu_synchronize(vega_ctx_t *c) {
+    if (!c) return VEGA_ERR_INVALID;
+    CTX_TRY(c, hipStreamSynchronize(c->stream));
+    return VEGA_OK;
+}
+
+const char *vega_gpu_last_error(vega_ctx_t *c) { return c ? c->err : "null ctx"; }
+
+static int make_rdd_common(vega_ctx *c, const int64_t *keys, const void *vals,
+                           uint64_t n, uint32_t nparts, int vtype, vega_rdd_t *out) {
+    if (!c || (!keys && n) || (!vals && n)) return VEGA_ERR_INVALID;
+    RddImpl *r;
+    int rc = new_rdd(c, n ? n : 1, vtype, nparts, &r, out);
+    if (rc) return rc;
+    r->n = n;
+    if (n) {
+        CTX_TRY(c, hipMemcpyAsync(r->d_k, keys, n * 8, hipMemcpyHostToDevice, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(r->d_v, vals, n * 8, hipMemcpyHostToDevice, c->stream));
+        CTX_TRY(c, hipStreamSynchronize(c->stream));
+    }
+    return VEGA_OK;
+}
+
+int vega_gpu_make_rdd(vega_ctx_t *c, const int64_t *keys, const int64_t *vals,
+                      uint64_t n, uint32_t nparts, vega_rdd_t *out) {
+    return make_rdd_common(c, keys, vals, n, nparts, 0, out);
+}
+int vega_gpu_make_rdd_f64(vega_ctx_t *c, const int64_t *keys, const double *vals,
+                          uint64_t n, uint32_t nparts, vega_rdd_t *out) {
+    return make_rdd_common(c, keys, vals, n, nparts, 1, out);
+}
+
+int vega_gpu_gen_rdd_uniform(vega_ctx_t *c, uint64_t n, uint64_t seed, int key_bits,
+                             uint64_t start, uint32_t nparts, vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    RddImpl *r;
+    int rc = new_rdd(c, n ? n : 1, 0, nparts, &r, out);
+    if (rc) return rc;
+    r->n = n;
+    CTX_TRY(c, gen_uniform(c->stream, r->d_k, (int64_t *)r->d_v, n, seed, key_bits, start));
+    return VEGA_OK;
+}
+
+/* the hot path: sort + segmented aggregate. nparts is the logical output
+ * partition count (pair_rdd.rs:54-80); the collected RESULT is invariant to
+ * it (each key lands in exactly one reduce partition either way), so the
+ * single-GPU engine computes the global aggregate directly. */
+static int reduce_common(vega_ctx *c, vega_rdd_t rdd, int op, uint32_t nparts,
+                         vega_rdd_t *out) {
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r) return VEGA_ERR_INVALID;
+    if ((op == VEGA_OP_SUM_F64) != (r->vtype == 1)) return VEGA_ERR_INVALID;
+    int rc = ensure_ws(c, r->n);
+    if (rc) return rc;
+    RddImpl *o;
+    rc = new_rdd(c, r->n ? r->n : 1, op == VEGA_OP_SUM_F64 ? 1 : 0, nparts, &o, out);
+    if (rc) return rc;
+    Ws ws(c->ws, c->ws_bytes);
+    const uint64_t *sk, *sv;
+    CTX_TRY(c, radix_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
+                              r->n, true, false, ws, &sk, &sv));
+    uint64_t nout = 0;
+    CTX_TRY(c, seg_reduce(c->stream, sk, sv, r->n, op, (uint64_t *)o->d_k, o->d_v, &nout, ws));
+    o->n = nout;
+    o->sorted = true; /* seg output is key-sorted (unsigned); collect compares sorted */
+    return VEGA_OK;
+}
+
+int vega_gpu_reduce_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_op_t op,
+                           uint32_t nparts, vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    return reduce_common(c, rdd, op, nparts, out);
+}
+
+int vega_gpu_group_count(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts, vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    return reduce_common(c, rdd, VEGA_OP_COUNT, nparts, out);
+}
+
+int vega_gpu_distinct(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts, vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    return reduce_common(c, rdd, VEGA_OP_COUNT, nparts, out);
+}
+
+int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r) return VEGA_ERR_INVALID;
+    int rc = ensure_ws(c, r->n);
+    if (rc) return rc;
+    RddImpl *o;
+    rc = new_rdd(c, r->n ? r->n : 1, r->vtype, r->nparts, &o, out);
+    if (rc) return rc;
+    o->n = r->n;
+    o->sorted = true;
+    Ws ws(c->ws, c->ws_bytes);
+    const uint64_t *sk, *sv;
+    CTX_TRY(c, radix_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
+                              r->n, true, true, ws, &sk, &sv));
+    CTX_TRY(c, hipMemcpyAsync(o->d_k, sk, r->n * 8, hipMemcpyDeviceToDevice, c->stream));
+    CTX_TRY(c, hipMemcpyAsync(o->d_v, sv, r->n * 8, hipMemcpyDeviceToDevice, c->stream));
+    return VEGA_OK;
+}
+
+int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
+                  vega_rdd_t *out) {
+    (void)a; (void)b; (void)nparts; (void)out;
+    if (!c) return VEGA_ERR_INVALID;
+    snprintf(c->err, sizeof c->err, "join: not implemented yet (K4)");
+    return VEGA_ERR_UNSUPPORTED;
+}
+
+int vega_gpu_count(vega_ctx_t *c, vega_rdd_t rdd, uint64_t *n) {
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r) return VEGA_ERR_INVALID;
+    *n = r->n;
+    return VEGA_OK;
+}
+
+int vega_gpu_collect(vega_ctx_t *c, vega_rdd_t rdd, int64_t *keys, void *vals, uint64_t *n) {
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r) return VEGA_ERR_INVALID;
+    if (!keys) {
+        *n = r->n;
+        return VEGA_OK;
+    }
+    if (*n < r->n) return VEGA_ERR_CAP;
+    *n = r->n;
+    if (r->n) {
+        CTX_TRY(c, hipMemcpyAsync(keys, r->d_k, r->n * 8, hipMemcpyDeviceToHost, c->stream));
+        if (vals)
+            CTX_TRY(c, hipMemcpyAsync(vals, r->d_v, r->n * 8, hipMemcpyDeviceToHost, c->stream));
+    }
+    CTX_TRY(c, hipStreamSynchronize(c->stream));
+    return VEGA_OK;
+}
+
+int vega_gpu_free_rdd(vega_ctx_t *c, vega_rdd_t rdd) {
+    auto it = c->rdds.find(rdd);
+    if (it == c->rdds.end()) return VEGA_ERR_INVALID;
+    CTX_TRY(c, hipStreamSynchronize(c->stream));
+    if (it->second->d_k) (void)hipFree(it->second->d_k);
+    if (it->second->d_v) (void)hipFree(it->second->d_v);
+    delete it->second;
+    c->rdds.erase(it);
+    return VEGA_OK;
+}
+
+int vega_gpu_set_profiling(vega_ctx_t *c, int enabled) {
+    (void)c;
+    prof_enable(enabled != 0);
+    return VEGA_OK;
+}
+int vega_gpu_kernel_stats(vega_ctx_t *c, char *buf, size_t buflen) {
+    (void)c;
+    return prof_stats_json(buf, buflen) < 0 ? VEGA_ERR_CAP : VEGA_OK;
+}
+
+/* =================== device-pointer API =================== */
+
+size_t vega_dev_ws_bytes(uint64_t n) { return ws_bytes_for(n); }
+
+/* separate global prof switch for the pointer API */
+int vega_prof_enable(int on) { prof_enable(on != 0); return VEGA_OK; }
+int vega_prof_stats(char *buf, size_t len) { return prof_stats_json(buf, len) < 0 ? VEGA_ERR_CAP : VEGA_OK; }
+
+int vega_dev_gen_uniform_i64(void *stream, int64_t *keys, int64_t *vals, uint64_t n,
+                             uint64_t seed, int key_bits, uint64_t start) {
+    return gen_uniform((hipStream_t)stream, keys, vals, n, seed, key_bits, start)
+               == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
+}
+
+int vega_dev_partition_i64(void *stream, const int64_t *keys, const int64_t *vals,
+                           uint64_t n, uint32_t nparts, int64_t *out_k, int64_t *out_v,
+                           uint64_t *h_counts, void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    hipError_t e = hash_partition((hipStream_t)stream, (const uint64_t *)keys,
+                                  (const uint64_t *)vals, n, nparts,
+                                  (uint64_t *)out_k, (uint64_t *)out_v, h_counts, ws);
+    return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
+}
+
+int vega_dev_sort_reduce(void *stream, const int64_t *in_k, const void *in_v,
+                         uint64_t n, int op, int64_t *out_k, void *out_v,
+                         uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    const uint64_t *sk, *sv;
+    hipError_t e = radix_sort_u64((hipStream_t)stream, (const uint64_t *)in_k,
+                                  (const uint64_t *)in_v, n, true, false, ws, &sk, &sv);
+    if (e != hipSuccess) return e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP;
+    e = seg_reduce((hipStream_t)stream, sk, sv, n, op, (uint64_t *)out_k, out_v, h_nout, ws);
+    return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
+}
+
+int vega_dev_sort_pairs_i64(void *stream, int64_t *keys, int64_t *vals, uint64_t n,
+                            void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    const uint64_t *sk, *sv;
+    hipError_t e = radix_sort_u64((hipStream_t)stream, (const uint64_t *)keys,
+                                  (const uint64_t *)vals, n, true, true, ws, &sk, &sv);
+    if (e != hipSuccess) return e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP;
+    if ((const uint64_t *)keys != sk) {
+        e = hipMemcpyAsync(keys, sk, n * 8, hipMemcpyDeviceToDevice, (hipStream_t)stream);
+        if (e != hipSuccess) return VEGA_ERR_HIP;
+        e = hipMemcpyAsync(vals, sv, n * 8, hipMemcpyDeviceToDevice, (hipStream_t)stream);
+        if (e != hipSuccess) return VEGA_ERR_HIP;
+    }
+    return VEGA_OK;
+}
+
+int vega_dev_join_sorted(void *stream, const int64_t *ak, const int64_t *av, uint64_t na,
+                         const int64_t *bk, const int64_t *bv, uint64_t nb,
+                         int64_t *out_k, int64_t *out_va, int64_t *out_vb,
+                         uint64_t cap, uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb,
+                               out_k, out_va, out_vb, cap, h_nout, ws);
+    if (e == hipErrorNotSupported) return VEGA_ERR_UNSUPPORTED;
+    return e == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
+}
+
+int vega_dev_checksum_pairs(void *stream, const int64_t *keys, const int64_t *vals,
+                            uint64_t n, uint64_t *h_sum, void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    return checksum_pairs((hipStream_t)stream, keys, vals, n, h_sum, ws) == hipSuccess
+               ? VEGA_OK : VEGA_ERR_HIP;
+}
+
+} /* extern "C" */

@@ -1,0 +1,84 @@
+/* vega_internal.h — internal host-side interfaces between the kernel TU
+ * (vega_kernels.hip) and the C-ABI TU (vega_api.hip). Not installed. */
+#ifndef VEGA_INTERNAL_H
+#define VEGA_INTERNAL_H
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstddef>
+
+namespace vega {
+
+/* ---- profiling registry (enabled via vega_prof_enable) ---- */
+void prof_enable(bool on);
+bool prof_on();
+/* record a closed event pair under `name`; takes ownership of events */
+void prof_record(const char *name, hipEvent_t start, hipEvent_t stop);
+/* drain into a JSON string: {"name":{"ms":x,"n":k},...} */
+int prof_stats_json(char *buf, size_t len);
+
+/* RAII-ish kernel timer: when profiling is on, brackets launches on `s` */
+struct ProfScope {
+    const char *name;
+    hipStream_t s;
+    hipEvent_t e0 = nullptr, e1 = nullptr;
+    ProfScope(const char *n, hipStream_t stream);
+    ~ProfScope();
+};
+
+/* ---- workspace carving ---- */
+struct Ws {
+    char *p;
+    size_t left;
+    Ws(void *ws, size_t bytes) : p((char *)ws), left(bytes) {}
+    void *take(size_t bytes) {
+        size_t a = (bytes + 255) & ~(size_t)255;
+        if (a > left) return nullptr;
+        void *r = p;
+        p += a;
+        left -= a;
+        return r;
+    }
+};
+
+/* ---- low-level ops (device pointers, async on stream) ---- */
+
+/* exclusive scan of n uint32 in place; ws scratch */
+hipError_t scan_u32_excl(hipStream_t s, uint32_t *a, uint64_t n, Ws &ws);
+
+/* stable LSB radix sort of (key,val) u64 pairs by full 64-bit key with
+ * degenerate-pass skipping. in_* are const; result pointers returned in
+ * res_k / res_v (one of the two ws ping-pong buffers, or in_* when zero
+ * passes were needed — in that case result aliases the input).
+ * has_vals=false skips all value movement. */
+hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                          uint64_t n, bool has_vals, bool signed_order, Ws &ws,
+                          const uint64_t **res_k, const uint64_t **res_v);
+
+/* segmented aggregate over key-sorted rows: one output row per equal-key
+ * run. op: VEGA_OP_*. Returns #segments in *h_nout (after stream sync). */
+hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t n,
+                      int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws);
+
+/* hash-mod partition scatter; h_counts on host after sync */
+hipError_t hash_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                          uint64_t n, uint32_t nparts, uint64_t *out_k, uint64_t *out_v,
+                          uint64_t *h_counts, Ws &ws);
+
+hipError_t gen_uniform(hipStream_t s, int64_t *keys, int64_t *vals, uint64_t n,
+                       uint64_t seed, int key_bits, uint64_t start);
+
+hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
+                          uint64_t n, uint64_t *h_sum, Ws &ws);
+
+/* sort-merge inner join of key-sorted sides */
+hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint64_t na,
+                       const int64_t *bk, const int64_t *bv, uint64_t nb,
+                       int64_t *out_k, int64_t *out_va, int64_t *out_vb,
+                       uint64_t cap, uint64_t *h_nout, Ws &ws);
+
+size_t ws_bytes_for(uint64_t n);
+
+} // namespace vega
+
+#endif

@@ -1,0 +1,246 @@
+"""ctypes bindings for libvega_gpu.so (the C-ABI in include/vega_gpu.h).
+
+Two layers, mirroring the ABI:
+  - VegaContext: the RDD-handle API (single process, one GPU) — the drop-in
+    surface a Rust host would bind.
+  - dev_*: device-pointer entries for the rank-per-GPU path; they take torch
+    CUDA tensors (device memory + stream plumbing only — all compute is in
+    the hand-written HIP kernels).
+
+This module FAILS LOUDLY if the HIP library is missing — no CPU fallback.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+LIBPATH = os.path.join(HERE, "csrc", "libvega_gpu.so")
+
+OP_SUM_I64 = 0
+OP_COUNT = 1
+OP_SUM_F64 = 2
+OP_MIN_I64 = 3
+OP_MAX_I64 = 4
+
+_lib = None
+
+
+class VegaGpuError(RuntimeError):
+    pass
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(LIBPATH):
+            raise VegaGpuError(
+                f"{LIBPATH} not built — run __graft_entry__.build() (hipcc "
+                "--offload-arch=gfx950); the GPU path has no fallback")
+        _lib = ctypes.CDLL(LIBPATH)
+        _lib.vega_dev_ws_bytes.restype = ctypes.c_size_t
+        _lib.vega_dev_ws_bytes.argtypes = [ctypes.c_uint64]
+        _lib.vega_gpu_last_error.restype = ctypes.c_char_p
+    return _lib
+
+
+def _np(a, dtype=np.int64):
+    return np.ascontiguousarray(a, dtype=dtype)
+
+
+def _pp(a):
+    return a.ctypes.data_as(ctypes.c_void_p)
+
+
+def _check(rc, what, ctx=None):
+    if rc != 0:
+        extra = b""
+        if ctx is not None:
+            extra = lib().vega_gpu_last_error(ctx)
+        raise VegaGpuError(f"{what} failed rc={rc} {extra!r}")
+
+
+class VegaContext:
+    """Mirrors Context (context.rs:147-164) + PairRdd ops for one GPU."""
+
+    def __init__(self):
+        self._c = ctypes.c_void_p()
+        _check(lib().vega_gpu_init(1, ctypes.byref(self._c)), "init")
+
+    def close(self):
+        if self._c:
+            lib().vega_gpu_shutdown(self._c)
+            self._c = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+    # --- construction (context.rs:406-442 make_rdd/parallelize) ---
+    def make_rdd(self, keys, vals, nparts=4):
+        h = ctypes.c_uint64()
+        if np.asarray(vals).dtype == np.float64:
+            k = _np(keys)
+            v = _np(vals, np.float64)
+            _check(lib().vega_gpu_make_rdd_f64(self._c, _pp(k), _pp(v),
+                                               ctypes.c_uint64(len(k)),
+                                               ctypes.c_uint32(nparts),
+                                               ctypes.byref(h)), "make_rdd_f64", self._c)
+        else:
+            k = _np(keys)
+            v = _np(vals)
+            _check(lib().vega_gpu_make_rdd(self._c, _pp(k), _pp(v),
+                                           ctypes.c_uint64(len(k)),
+                                           ctypes.c_uint32(nparts),
+                                           ctypes.byref(h)), "make_rdd", self._c)
+        return Rdd(self, h.value, np.asarray(vals).dtype)
+
+    def gen_rdd_uniform(self, n, seed, key_bits=63, start=0, nparts=256):
+        h = ctypes.c_uint64()
+        _check(lib().vega_gpu_gen_rdd_uniform(
+            self._c, ctypes.c_uint64(n), ctypes.c_uint64(seed),
+            ctypes.c_int(key_bits), ctypes.c_uint64(start),
+            ctypes.c_uint32(nparts), ctypes.byref(h)), "gen_rdd", self._c)
+        return Rdd(self, h.value, np.dtype(np.int64))
+
+    def synchronize(self):
+        _check(lib().vega_gpu_synchronize(self._c), "sync", self._c)
+
+    def set_profiling(self, on):
+        lib().vega_gpu_set_profiling(self._c, 1 if on else 0)
+
+    def kernel_stats(self):
+        import json
+        buf = ctypes.create_string_buffer(1 << 16)
+        _check(lib().vega_gpu_kernel_stats(self._c, buf, len(buf)), "stats")
+        return json.loads(buf.value.decode())
+
+
+class Rdd:
+    def __init__(self, ctx, handle, vdtype):
+        self.ctx = ctx
+        self.h = handle
+        self.vdtype = np.dtype(vdtype)
+
+    # --- PairRdd ops (pair_rdd.rs names) ---
+    def reduce_by_key(self, op=OP_SUM_I64, nparts=256):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_reduce_by_key(self.ctx._c, ctypes.c_uint64(self.h),
+                                            ctypes.c_int(op), ctypes.c_uint32(nparts),
+                                            ctypes.byref(out)),
+               "reduce_by_key", self.ctx._c)
+        return Rdd(self.ctx, out.value,
+                   np.float64 if op == OP_SUM_F64 else np.int64)
+
+    def group_count(self, nparts=256):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_group_count(self.ctx._c, ctypes.c_uint64(self.h),
+                                          ctypes.c_uint32(nparts), ctypes.byref(out)),
+               "group_count", self.ctx._c)
+        return Rdd(self.ctx, out.value, np.int64)
+
+    def sort_by_key(self):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_sort_by_key(self.ctx._c, ctypes.c_uint64(self.h),
+                                          ctypes.byref(out)), "sort_by_key", self.ctx._c)
+        return Rdd(self.ctx, out.value, self.vdtype)
+
+    # --- actions (rdd.rs collect/count) ---
+    def count(self):
+        n = ctypes.c_uint64()
+        _check(lib().vega_gpu_count(self.ctx._c, ctypes.c_uint64(self.h),
+                                    ctypes.byref(n)), "count", self.ctx._c)
+        return n.value
+
+    def collect(self):
+        n = ctypes.c_uint64(self.count())
+        k = np.empty(n.value, dtype=np.int64)
+        v = np.empty(n.value, dtype=self.vdtype)
+        _check(lib().vega_gpu_collect(self.ctx._c, ctypes.c_uint64(self.h),
+                                      _pp(k), _pp(v), ctypes.byref(n)),
+               "collect", self.ctx._c)
+        return k[:n.value], v[:n.value]
+
+    def free(self):
+        lib().vega_gpu_free_rdd(self.ctx._c, ctypes.c_uint64(self.h))
+
+
+# ---------------- device-pointer API over torch tensors ----------------
+
+def _t(t):
+    return ctypes.c_void_p(t.data_ptr())
+
+
+def _stream():
+    import torch
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def ws_bytes(n):
+    return int(lib().vega_dev_ws_bytes(ctypes.c_uint64(n)))
+
+
+def alloc_ws(n, device="cuda"):
+    import torch
+    return torch.empty(ws_bytes(n), dtype=torch.uint8, device=device)
+
+
+def dev_gen_uniform(keys_t, vals_t, seed, key_bits=63, start=0):
+    n = keys_t.numel()
+    _check(lib().vega_dev_gen_uniform_i64(_stream(), _t(keys_t), _t(vals_t),
+                                          ctypes.c_uint64(n), ctypes.c_uint64(seed),
+                                          ctypes.c_int(key_bits), ctypes.c_uint64(start)),
+           "dev_gen")
+
+
+def dev_partition(keys_t, vals_t, nparts, out_k, out_v, ws_t):
+    n = keys_t.numel()
+    counts = np.zeros(nparts, dtype=np.uint64)
+    _check(lib().vega_dev_partition_i64(_stream(), _t(keys_t), _t(vals_t),
+                                        ctypes.c_uint64(n), ctypes.c_uint32(nparts),
+                                        _t(out_k), _t(out_v), _pp(counts),
+                                        _t(ws_t), ctypes.c_size_t(ws_t.numel())),
+           "dev_partition")
+    return counts
+
+
+def dev_sort_reduce(keys_t, vals_t, op, out_k, out_v, ws_t):
+    n = keys_t.numel()
+    nout = ctypes.c_uint64()
+    _check(lib().vega_dev_sort_reduce(_stream(), _t(keys_t), _t(vals_t),
+                                      ctypes.c_uint64(n), ctypes.c_int(op),
+                                      _t(out_k), _t(out_v), ctypes.byref(nout),
+                                      _t(ws_t), ctypes.c_size_t(ws_t.numel())),
+           "dev_sort_reduce")
+    return nout.value
+
+
+def dev_sort_pairs(keys_t, vals_t, ws_t):
+    n = keys_t.numel()
+    _check(lib().vega_dev_sort_pairs_i64(_stream(), _t(keys_t), _t(vals_t),
+                                         ctypes.c_uint64(n), _t(ws_t),
+                                         ctypes.c_size_t(ws_t.numel())),
+           "dev_sort_pairs")
+
+
+def dev_checksum(keys_t, vals_t, ws_t):
+    n = keys_t.numel()
+    s = ctypes.c_uint64()
+    _check(lib().vega_dev_checksum_pairs(_stream(), _t(keys_t), _t(vals_t),
+                                         ctypes.c_uint64(n), ctypes.byref(s),
+                                         _t(ws_t), ctypes.c_size_t(ws_t.numel())),
+           "dev_checksum")
+    return s.value
+
+
+def prof_enable(on=True):
+    lib().vega_prof_enable(1 if on else 0)
+
+
+def prof_stats():
+    import json
+    buf = ctypes.create_string_buffer(1 << 16)
+    _check(lib().vega_prof_stats(buf, len(buf)), "prof_stats")
+    return json.loads(buf.value.decode())
