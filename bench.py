@@ -1,0 +1,233 @@
+#!/usr/bin/env python3
+"""bench.py — the driver-contract benchmark for the MI355X vega engine.
+
+Measures BASELINE.json's metric: rows/sec (whole node) reduce_by_key i64->i64
+on the C1 workload (1e9 (i64,i64) rows, uniform keys in [0,2^63), 256 logical
+partitions). A "step" is one reduce_by_key pass over the batch already
+resident in HBM (H2D/generation is untimed; collect is untimed — the
+PCIe-inclusive rate is discussed in DESIGN.md).
+
+  python bench.py --gpus N --steps K --warmup W
+  (N>1 is launched by the driver via torch.distributed.run, one rank per GPU;
+   ranks read RANK/LOCAL_RANK/WORLD_SIZE from the env. Scaling is WEAK: each
+   rank processes its own --rows shard of the global stream; the exchange is
+   the counts-all-to-all + all-to-all-v over RCCL/xGMI.)
+
+Rank 0 prints ONE JSON line per the contract, plus:
+  roofline     — dominant kernel (radix_scatter) achieved GB/s vs the 8 TB/s
+                 HBM peak, measured with HIP events on the launch stream
+                 during the timed region (algorithmic bytes: DESIGN.md §Roofline)
+  cpu_baseline — the CPU oracle (vega's own bucketed-HashMap algorithm,
+                 oracle/oracle.c, OpenMP over partitions) timed on this box's
+                 host cores on a bounded sample (N=1 rank 0 only)
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+METRIC = ("rows/sec (whole node) reduce_by_key i64→i64, 1e9 rows, "
+          "1/2/4/8 GPU")
+WORKLOAD = ("C1: 1e9 (i64,i64) uniform keys in [0,2^63), reduce_by_key(sum), "
+            "256 partitions (BASELINE.json configs[1]); weak scaling: "
+            "--rows per GPU")
+HBM_PEAK_GBS = 8000.0          # spec peak, MI355X_MICROARCH.md
+SCATTER_ALGO_BYTES_PER_ROW = 32  # read (k,v) 16 B + write (k,v) 16 B per pass
+
+
+def log(msg):
+    print(msg, file=sys.stderr, flush=True)
+
+
+def cpu_baseline_leg(rows, key_bits, seed):
+    """Time the CPU oracle (kind=port: restatement of dependency.rs:176-223 +
+    shuffled_rdd.rs:153-169) on a bounded sample of the same workload."""
+    sys.path.insert(0, os.path.join(ROOT, "tests"))
+    import oracle_ctypes as oc
+    from vega_amd import datagen
+    cores = os.cpu_count()
+    pilot = min(rows, 2_000_000)
+    k, v = datagen.uniform_pairs(seed, pilot, key_bits=key_bits)
+    t0 = time.perf_counter()
+    oc.reduce_by_key_i64(k, v, 256, 256)
+    dt = time.perf_counter() - t0
+    rate = pilot / dt
+    sample = int(min(rows, max(pilot, min(rate * 15.0, 400_000_000))))
+    k, v = datagen.uniform_pairs(seed, sample, key_bits=key_bits)
+    t0 = time.perf_counter()
+    oc.reduce_by_key_i64(k, v, 256, 256)
+    dt = time.perf_counter() - t0
+    return {
+        "value": sample / dt,
+        "unit": "rows/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{sample} rows of the same uniform-key stream "
+                  f"({dt:.1f}s on {cores} host cores, OpenMP)",
+    }
+
+
+def read_traffic_calibration():
+    """Optional: measured per-launch HBM bytes for the dominant kernel from a
+    committed rocprofv3 PMC calibration (profiles/traffic_calib.json)."""
+    p = os.path.join(ROOT, "profiles", "traffic_calib.json")
+    if os.path.exists(p):
+        try:
+            return json.load(open(p)).get("radix_scatter_bytes_per_launch")
+        except Exception:
+            return None
+    return None
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--rows", type=int, default=1_000_000_000,
+                    help="rows per GPU (weak scaling)")
+    ap.add_argument("--key-bits", type=int, default=63)
+    ap.add_argument("--seed", type=int, default=0xC0FFEE + 1)  # config C1
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    from vega_amd import gpu, shuffle
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world > 1:
+        import torch.distributed as dist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+    else:
+        torch.cuda.set_device(0)
+    n_gpus = world if world > 1 else args.gpus
+    assert n_gpus == world or world == 1, "launch N>1 via torch.distributed.run"
+
+    rows = args.rows
+    dev = torch.device("cuda")
+
+    # ---- setup (untimed): generate the shard resident in HBM ----
+    k = torch.empty(rows, dtype=torch.int64, device=dev)
+    v = torch.empty(rows, dtype=torch.int64, device=dev)
+    gpu.dev_gen_uniform(k, v, seed=args.seed, key_bits=args.key_bits,
+                        start=rank * rows)
+    slack = 1.10 if world > 1 else 1.0
+    cap = int(rows * slack) + 1024
+    ws = gpu.alloc_ws(cap)
+    out_k = torch.empty(cap, dtype=torch.int64, device=dev)
+    out_v = torch.empty(cap, dtype=torch.int64, device=dev)
+    if world > 1:
+        pk = torch.empty(rows, dtype=torch.int64, device=dev)
+        pv = torch.empty(rows, dtype=torch.int64, device=dev)
+
+    nout = None
+
+    def step():
+        nonlocal nout
+        if world == 1:
+            nout = gpu.dev_sort_reduce(k, v, gpu.OP_SUM_I64, out_k, out_v, ws)
+        else:
+            counts = gpu.dev_partition(k, v, world, pk, pv, ws)
+            rk, rv = shuffle.all_to_all_kv(pk, pv, counts.astype(np.int64).tolist())
+            if rk.numel() > cap:
+                raise RuntimeError(f"recv shard {rk.numel()} > cap {cap}")
+            nout = gpu.dev_sort_reduce(rk, rv, gpu.OP_SUM_I64, out_k, out_v, ws)
+
+    def barrier_sync():
+        torch.cuda.synchronize()
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+            torch.cuda.synchronize()
+
+    log(f"[rank {rank}] warmup x{args.warmup} (rows={rows})")
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+
+    profiling = (rank == 0 and world == 1)
+    if profiling:
+        gpu.prof_enable(True)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_rows = rows * n_gpus
+    value = total_rows / (elapsed / args.steps)
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    roofline = None
+    if profiling:
+        stats = gpu.prof_stats()
+        sc = stats.get("radix_scatter")
+        if sc and sc["n"] > 0:
+            avg_ms = sc["ms"] / sc["n"]
+            algo_gb = SCATTER_ALGO_BYTES_PER_ROW * rows / 1e9
+            achieved = algo_gb / (avg_ms / 1000.0)
+            roofline = {
+                "bound": "hbm",
+                "achieved": round(achieved, 1),
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": round(achieved / HBM_PEAK_GBS, 4),
+                "traffic": read_traffic_calibration(),
+                "kernel": "radix_scatter",
+                "launches": sc["n"],
+                "avg_launch_ms": round(avg_ms, 3),
+            }
+        log(f"[rank 0] kernel stats: {json.dumps(stats)}")
+
+    if rank == 0:
+        out = {
+            "metric": METRIC,
+            "value": round(value, 1),
+            "unit": "rows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": WORKLOAD,
+                "rows_per_gpu": rows,
+                "key_bits": args.key_bits,
+                "nparts": 256,
+                "distinct_keys_out": int(nout) if world == 1 else None,
+            },
+            "roofline": roofline,
+            "cpu_baseline": (cpu_baseline_leg(rows, args.key_bits, args.seed)
+                             if (world == 1 and not args.no_cpu_baseline) else None),
+        }
+        print(json.dumps(out), flush=True)
+
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -16,6 +16,7 @@ import numpy as np
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 LIBPATH = os.path.join(HERE, "csrc", "libvega_gpu.so")
+LIBPATH_TORCH = os.path.join(HERE, "csrc", "libvega_gpu_torch.so")
 
 OP_SUM_I64 = 0
 OP_COUNT = 1
@@ -33,11 +34,21 @@ class VegaGpuError(RuntimeError):
 def lib():
     global _lib
     if _lib is None:
-        if not os.path.exists(LIBPATH):
+        # Prefer the torch-runtime-linked variant and load torch FIRST, so
+        # exactly one HIP runtime (torch's bundled one) lives in the process;
+        # torch tensors/streams and our kernels then share it.
+        path = LIBPATH
+        if os.path.exists(LIBPATH_TORCH):
+            try:
+                import torch  # noqa: F401  (loads its libamdhip64.so)
+                path = LIBPATH_TORCH
+            except ImportError:
+                pass
+        if not os.path.exists(path):
             raise VegaGpuError(
-                f"{LIBPATH} not built — run __graft_entry__.build() (hipcc "
+                f"{path} not built — run __graft_entry__.build() (hipcc "
                 "--offload-arch=gfx950); the GPU path has no fallback")
-        _lib = ctypes.CDLL(LIBPATH)
+        _lib = ctypes.CDLL(path)
         _lib.vega_dev_ws_bytes.restype = ctypes.c_size_t
         _lib.vega_dev_ws_bytes.argtypes = [ctypes.c_uint64]
         _lib.vega_gpu_last_error.restype = ctypes.c_char_p

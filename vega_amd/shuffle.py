@@ -1,0 +1,62 @@
+"""Rank-per-GPU shuffle exchange (replaces the reference's HTTP-pull shuffle
+plane: shuffle_manager.rs:86-118 server + shuffle_fetcher.rs:61-90 client +
+map_output_tracker.rs URI registry).
+
+MI355X-native plan: one process per GPU over
+torch.distributed; bucket b of every map shard is owned by rank b; the
+exchange is ONE counts all-to-all (replacing the MapOutputTracker) followed
+by one all-to-all-v of the bucket payloads (RCCL over xGMI on GPUs; gloo on
+CPU for tests — same code path).
+"""
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def hash_u64_np(keys):
+    """vectorized splitmix64 (bit-identical to vega_common.h vega_hash_u64)"""
+    x = np.asarray(keys).astype(np.int64).view(np.uint64).copy()
+    with np.errstate(over="ignore"):
+        x = x + np.uint64(0x9E3779B97F4A7C15)
+        x = (x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        x = (x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+        x = x ^ (x >> np.uint64(31))
+    return x
+
+
+def bucket_of_np(keys, nparts):
+    return (hash_u64_np(keys) % np.uint64(nparts)).astype(np.int64)
+
+
+def partition_cpu(k, v, nparts):
+    """CPU reference partition (tests / gloo path): bucket-contiguous reorder
+    + per-bucket counts. Stable within buckets (row order preserved), matching
+    the GPU k_scatter's stable counting scatter."""
+    k = np.asarray(k, dtype=np.int64)
+    v = np.asarray(v, dtype=np.int64)
+    b = bucket_of_np(k, nparts)
+    order = np.argsort(b, kind="stable")
+    counts = np.bincount(b, minlength=nparts).astype(np.int64)
+    return k[order], v[order], counts
+
+
+def all_to_all_kv(send_k, send_v, send_counts, group=None):
+    """Exchange bucket-contiguous (k, v) rows: counts all-to-all, then
+    payload all-to-all-v. Tensors stay on their device (CUDA -> RCCL/xGMI,
+    CPU -> gloo)."""
+    dev = send_k.device
+    sc = torch.as_tensor(send_counts, dtype=torch.int64, device=dev)
+    rc = torch.empty_like(sc)
+    dist.all_to_all_single(rc, sc, group=group)
+    in_splits = [int(x) for x in sc.tolist()]
+    out_splits = [int(x) for x in rc.tolist()]
+    nrecv = sum(out_splits)
+    recv_k = torch.empty(nrecv, dtype=send_k.dtype, device=dev)
+    recv_v = torch.empty(nrecv, dtype=send_v.dtype, device=dev)
+    dist.all_to_all_single(recv_k, send_k[:sum(in_splits)],
+                           output_split_sizes=out_splits,
+                           input_split_sizes=in_splits, group=group)
+    dist.all_to_all_single(recv_v, send_v[:sum(in_splits)],
+                           output_split_sizes=out_splits,
+                           input_split_sizes=in_splits, group=group)
+    return recv_k, recv_v
