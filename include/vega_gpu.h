@@ -101,6 +101,9 @@ int vega_gpu_count(vega_ctx_t *ctx, vega_rdd_t rdd, uint64_t *n);
 /* collect (rdd.rs:420-434): D2H of the rows. Call with keys==NULL to query n. */
 int vega_gpu_collect(vega_ctx_t *ctx, vega_rdd_t rdd, int64_t *keys, void *vals,
                      uint64_t *n);
+/* collect of a join result (K,(V,W)) — pair_rdd.rs:104-121's output shape */
+int vega_gpu_collect_join(vega_ctx_t *ctx, vega_rdd_t rdd, int64_t *keys,
+                          int64_t *va, int64_t *vb, uint64_t *n);
 int vega_gpu_free_rdd(vega_ctx_t *ctx, vega_rdd_t rdd);
 
 /* ---------------- profiling ---------------- */

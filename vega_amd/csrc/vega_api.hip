@@ -24,6 +24,7 @@ struct RddImpl {
     int vtype = 0; /* 0 = i64 vals, 1 = f64 vals */
     int64_t *d_k = nullptr;
     void *d_v = nullptr;
+    void *d_v2 = nullptr; /* second value column (join output (K,(V,W))) */
     uint64_t alloc_rows = 0;
     uint32_t nparts = 1;
     bool sorted = false;
@@ -104,6 +105,7 @@ int vega_gpu_shutdown(vega_ctx_t *c) {
     for (auto &kv : c->rdds) {
         if (kv.second->d_k) (void)hipFree(kv.second->d_k);
         if (kv.second->d_v) (void)hipFree(kv.second->d_v);
+        if (kv.second->d_v2) (void)hipFree(kv.second->d_v2);
         delete kv.second;
     }
     if (c->ws) (void)hipFree(c->ws);
@@ -216,12 +218,69 @@ int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
     return VEGA_OK;
 }
 
+/* inner join (pair_rdd.rs:104-121 via cogroup co_grouped_rdd.rs:206-249):
+ * sort both sides, then sort-merge count + emit (K4). */
 int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
                   vega_rdd_t *out) {
-    (void)a; (void)b; (void)nparts; (void)out;
     if (!c) return VEGA_ERR_INVALID;
-    snprintf(c->err, sizeof c->err, "join: not implemented yet (K4)");
-    return VEGA_ERR_UNSUPPORTED;
+    RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
+    if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
+    vega_rdd_t ha = 0, hb = 0;
+    int rc = vega_gpu_sort_by_key(c, a, &ha);
+    if (rc) return rc;
+    rc = vega_gpu_sort_by_key(c, b, &hb);
+    if (rc) { vega_gpu_free_rdd(c, ha); return rc; }
+    RddImpl *sa = get_rdd(c, ha), *sb = get_rdd(c, hb);
+    uint64_t total = 0;
+    {
+        Ws ws(c->ws, c->ws_bytes);
+        hipError_t e = join_sorted(c->stream, sa->d_k, (const int64_t *)sa->d_v, sa->n,
+                                   sb->d_k, (const int64_t *)sb->d_v, sb->n,
+                                   nullptr, nullptr, nullptr, 0, &total, ws);
+        if (e != hipSuccess) {
+            vega_gpu_free_rdd(c, ha); vega_gpu_free_rdd(c, hb);
+            snprintf(c->err, sizeof c->err, "join count: %s", hipGetErrorString(e));
+            return VEGA_ERR_HIP;
+        }
+    }
+    RddImpl *o;
+    rc = new_rdd(c, total ? total : 1, 0, nparts, &o, out);
+    if (rc) { vega_gpu_free_rdd(c, ha); vega_gpu_free_rdd(c, hb); return rc; }
+    CTX_TRY(c, hipMalloc(&o->d_v2, (total ? total : 1) * 8));
+    o->n = total;
+    {
+        Ws ws(c->ws, c->ws_bytes);
+        uint64_t n2 = 0;
+        hipError_t e = join_sorted(c->stream, sa->d_k, (const int64_t *)sa->d_v, sa->n,
+                                   sb->d_k, (const int64_t *)sb->d_v, sb->n,
+                                   o->d_k, (int64_t *)o->d_v, (int64_t *)o->d_v2,
+                                   total, &n2, ws);
+        vega_gpu_free_rdd(c, ha);
+        vega_gpu_free_rdd(c, hb);
+        if (e != hipSuccess || n2 != total) {
+            snprintf(c->err, sizeof c->err, "join emit: %s (n2=%llu total=%llu)",
+                     hipGetErrorString(e), (unsigned long long)n2,
+                     (unsigned long long)total);
+            return VEGA_ERR_HIP;
+        }
+    }
+    return VEGA_OK;
+}
+
+int vega_gpu_collect_join(vega_ctx_t *c, vega_rdd_t rdd, int64_t *keys,
+                          int64_t *va, int64_t *vb, uint64_t *n) {
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r) return VEGA_ERR_INVALID;
+    if (!keys) { *n = r->n; return VEGA_OK; }
+    if (*n < r->n || !r->d_v2) return VEGA_ERR_CAP;
+    *n = r->n;
+    if (r->n) {
+        CTX_TRY(c, hipMemcpyAsync(keys, r->d_k, r->n * 8, hipMemcpyDeviceToHost, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(va, r->d_v, r->n * 8, hipMemcpyDeviceToHost, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(vb, r->d_v2, r->n * 8, hipMemcpyDeviceToHost, c->stream));
+    }
+    CTX_TRY(c, hipStreamSynchronize(c->stream));
+    return VEGA_OK;
 }
 
 int vega_gpu_count(vega_ctx_t *c, vega_rdd_t rdd, uint64_t *n) {
@@ -255,6 +314,7 @@ int vega_gpu_free_rdd(vega_ctx_t *c, vega_rdd_t rdd) {
     CTX_TRY(c, hipStreamSynchronize(c->stream));
     if (it->second->d_k) (void)hipFree(it->second->d_k);
     if (it->second->d_v) (void)hipFree(it->second->d_v);
+    if (it->second->d_v2) (void)hipFree(it->second->d_v2);
     delete it->second;
     c->rdds.erase(it);
     return VEGA_OK;

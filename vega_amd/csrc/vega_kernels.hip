@@ -696,11 +696,89 @@ size_t ws_bytes_for(uint64_t n) {
     return b;
 }
 
-/* join_sorted: implemented in a later milestone this round */
-hipError_t join_sorted(hipStream_t, const int64_t *, const int64_t *, uint64_t,
-                       const int64_t *, const int64_t *, uint64_t,
-                       int64_t *, int64_t *, int64_t *, uint64_t, uint64_t *, Ws &) {
-    return hipErrorNotSupported;
+/* ------------------------------------------------------------------ */
+/* K4: sort-merge inner join over key-sorted sides
+ * (replaces co_grouped_rdd.rs:206-249's HashMap-of-vecs + the cross-product
+ * flat_map_values of pair_rdd.rs:109-115). Per A row: binary search the
+ * equal-key run in B (log2 nb probes; upper tree levels stay in L2/L3),
+ * count + base, exclusive scan, then emit the cross product. */
+
+__global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
+                             uint64_t nb, uint32_t *counts, uint32_t *b_lo) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < na; i += stride) {
+        int64_t k = ak[i];
+        /* lower bound */
+        uint64_t lo = 0, hi = nb;
+        while (lo < hi) {
+            uint64_t m = (lo + hi) >> 1;
+            if (bk[m] < k) lo = m + 1; else hi = m;
+        }
+        uint64_t lb = lo;
+        /* upper bound */
+        hi = nb;
+        while (lo < hi) {
+            uint64_t m = (lo + hi) >> 1;
+            if (bk[m] <= k) lo = m + 1; else hi = m;
+        }
+        counts[i] = (uint32_t)(lo - lb);
+        b_lo[i] = (uint32_t)lb;
+    }
+}
+
+__global__ void k_join_emit(const int64_t *ak, const int64_t *av, uint64_t na,
+                            const int64_t *bv, const uint32_t *scan,
+                            const uint32_t *b_lo, int64_t *out_k,
+                            int64_t *out_va, int64_t *out_vb, uint64_t cap) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < na; i += stride) {
+        uint64_t base = scan[i];
+        uint32_t c = scan[i + 1] - (uint32_t)base;
+        int64_t k = ak[i], va = av[i];
+        uint32_t lb = b_lo[i];
+        for (uint32_t j = 0; j < c; ++j) {
+            uint64_t o = base + j;
+            if (o >= cap) break;
+            out_k[o] = k;
+            out_va[o] = va;
+            out_vb[o] = bv[lb + j];
+        }
+    }
+}
+
+hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint64_t na,
+                       const int64_t *bk, const int64_t *bv, uint64_t nb,
+                       int64_t *out_k, int64_t *out_va, int64_t *out_vb,
+                       uint64_t cap, uint64_t *h_nout, Ws &ws) {
+    if (na == 0 || nb == 0) { *h_nout = 0; return hipSuccess; }
+    uint32_t *counts = (uint32_t *)ws.take((na + 1) * 4);
+    uint32_t *b_lo = (uint32_t *)ws.take(na * 4);
+    if (!counts || !b_lo) return hipErrorOutOfMemory;
+    uint32_t nb_grid = nblocks_for(na);
+    uint32_t gb = nb_grid < 2048 ? nb_grid : 2048;
+    {
+        ProfScope ps("join_count", s);
+        hipLaunchKernelGGL(k_join_count, dim3(gb), dim3(BLOCK), 0, s, ak, na, bk, nb, counts, b_lo);
+        HIP_TRY(hipGetLastError());
+    }
+    HIP_TRY(hipMemsetAsync(counts + na, 0, 4, s));
+    {
+        Ws w2 = ws;
+        HIP_TRY(scan_u32_excl(s, counts, na + 1, w2));
+    }
+    uint32_t total = 0;
+    HIP_TRY(hipMemcpyAsync(&total, counts + na, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    *h_nout = total;
+    if (!out_k) return hipSuccess; /* count-only query */
+    if (total > cap) return hipErrorInvalidValue;
+    {
+        ProfScope ps("join_emit", s);
+        hipLaunchKernelGGL(k_join_emit, dim3(gb), dim3(BLOCK), 0, s, ak, av, na, bv,
+                           counts, b_lo, out_k, out_va, out_vb, cap);
+        HIP_TRY(hipGetLastError());
+    }
+    return hipSuccess;
 }
 
 } // namespace vega

@@ -158,6 +158,25 @@ class Rdd:
                                           ctypes.byref(out)), "sort_by_key", self.ctx._c)
         return Rdd(self.ctx, out.value, self.vdtype)
 
+    def join(self, other, nparts=256):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_join(self.ctx._c, ctypes.c_uint64(self.h),
+                                   ctypes.c_uint64(other.h), ctypes.c_uint32(nparts),
+                                   ctypes.byref(out)), "join", self.ctx._c)
+        r = Rdd(self.ctx, out.value, np.int64)
+        r.is_join = True
+        return r
+
+    def collect_join(self):
+        n = ctypes.c_uint64(self.count())
+        k = np.empty(n.value, dtype=np.int64)
+        va = np.empty(n.value, dtype=np.int64)
+        vb = np.empty(n.value, dtype=np.int64)
+        _check(lib().vega_gpu_collect_join(self.ctx._c, ctypes.c_uint64(self.h),
+                                           _pp(k), _pp(va), _pp(vb), ctypes.byref(n)),
+               "collect_join", self.ctx._c)
+        return k[:n.value], va[:n.value], vb[:n.value]
+
     # --- actions (rdd.rs collect/count) ---
     def count(self):
         n = ctypes.c_uint64()
