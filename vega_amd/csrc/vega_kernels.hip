@@ -248,9 +248,12 @@ hipError_t scan_u32_excl(hipStream_t s, uint32_t *a, uint64_t n, Ws &ws) {
 /* ------------------------------------------------------------------ */
 /* per-(digit, block) histogram                                        */
 
+/* writes the digit-major matrix bh[d][b] (for the device-wide scan) AND a
+ * block-major raw copy raw[b][d] (re-read by k_scatter for its local digit
+ * starts — saves an 8 B/row key re-read there) */
 template <class DF>
 __global__ void k_block_hist(const uint64_t *keys, uint64_t n, uint32_t nblocks,
-                             uint32_t *bh, DF df) {
+                             uint32_t *bh, uint32_t *raw, DF df) {
     __shared__ uint32_t h[256];
     for (int i = threadIdx.x; i < 256; i += BLOCK) h[i] = 0;
     __syncthreads();
@@ -261,8 +264,10 @@ __global__ void k_block_hist(const uint64_t *keys, uint64_t n, uint32_t nblocks,
         if (idx < n) atomicAdd(&h[df(keys[idx])], 1u);
     }
     __syncthreads();
-    for (int d = threadIdx.x; d < 256; d += BLOCK)
+    for (int d = threadIdx.x; d < 256; d += BLOCK) {
         bh[(uint64_t)d * nblocks + blockIdx.x] = h[d];
+        raw[(uint64_t)blockIdx.x * 256 + d] = h[d];
+    }
 }
 
 /* all 8 byte-position histograms in one pass (radix pass skipping) */
@@ -286,36 +291,34 @@ __global__ void k_hist8(const uint64_t *keys, uint64_t n, uint32_t *h8) {
 /* ------------------------------------------------------------------ */
 /* rank-and-scatter: stable counting scatter of one tile               */
 
+/* Stable counting scatter of one 4096-row tile.
+ * Ranking: each wave ranks its own contiguous 1024-row chunk over 16
+ * wave-synchronous rounds (ballot match, one LDS atomicAdd per distinct
+ * digit per round on the wave's PRIVATE counters) — no block barriers in the
+ * ranking loop; cross-wave offsets and the tile reorder take 3 barriers
+ * total. Local digit starts come from the precomputed raw block histogram
+ * (k_block_hist) instead of re-reading keys. */
 template <class DF, bool HAS_VALS>
 __global__ __launch_bounds__(BLOCK) void k_scatter(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
-    const uint32_t *bh_scanned, uint64_t *out_k, uint64_t *out_v, DF df) {
+    const uint32_t *bh_scanned, const uint32_t *raw_hist,
+    uint64_t *out_k, uint64_t *out_v, DF df) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *sk = (uint64_t *)smem;                              /* TILE u64 */
     uint64_t *sv = HAS_VALS ? sk + TILE : nullptr;                /* TILE u64 */
-    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 2 : 1) * TILE * 8); /* 256 */
-    uint32_t *run = hist + 256;                                   /* 256 */
-    uint32_t *cnt = run + 256;                                    /* 4*256 */
-    uint32_t *wsc = cnt + 4 * 256;                                /* 4 */
+    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 2 : 1) * TILE * 8); /* 256 local starts */
+    uint32_t *whist = hist + 256;                                 /* 4*256 per-wave counters */
+    uint32_t *wsc = whist + 4 * 256;                              /* 4 */
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t tbase = (uint64_t)blockIdx.x * TILE;
     const uint32_t tile_n = (uint32_t)((n - tbase < TILE) ? (n - tbase) : TILE);
+    const uint64_t lower = ((uint64_t)1 << lane) - 1;
 
-    for (int i = t; i < 256; i += BLOCK) { hist[i] = 0; run[i] = 0; }
-    __syncthreads();
-
-    /* pass A: block digit histogram (coalesced key reads) */
-#pragma unroll
-    for (int j = 0; j < IPT; ++j) {
-        uint64_t idx = tbase + (uint64_t)j * BLOCK + t;
-        if (idx < n) atomicAdd(&hist[df(in_k[idx])], 1u);
-    }
-    __syncthreads();
-
-    /* exclusive scan of hist -> local start of each digit in the tile */
+    for (int i = t; i < 4 * 256; i += BLOCK) whist[i] = 0;
+    /* local digit starts: exclusive scan of this block's raw histogram */
     {
-        uint32_t h = hist[t];
+        uint32_t h = raw_hist[(uint64_t)blockIdx.x * 256 + t];
         uint32_t inc = h;
         for (int off = 1; off < 64; off <<= 1) {
             uint32_t u = __shfl_up(inc, off);
@@ -325,17 +328,18 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
         __syncthreads();
         uint32_t excl = inc - h;
         for (int i = 0; i < w; ++i) excl += wsc[i];
-        __syncthreads();
         hist[t] = excl;
     }
-    __syncthreads();
+    /* no barrier needed before ranking: each wave only touches its own
+     * whist row, and hist[] is re-read only after the next barrier */
 
-    /* group loop: 16 groups of 256 items in tile order; stable rank via
-     * wave ballot match + cross-wave LDS counts */
-    for (int g = 0; g < IPT; ++g) {
-        for (int i = t; i < 4 * 256; i += BLOCK) cnt[i] = 0;
-        __syncthreads();
-        uint64_t idx = tbase + (uint64_t)g * BLOCK + t;
+    /* ranking: wave w ranks rows [w*1024, w*1024+1024) in 16 rounds */
+    uint64_t kk[IPT], vv[IPT];
+    uint32_t rank[IPT];
+    uint16_t dd[IPT];
+#pragma unroll
+    for (int r = 0; r < IPT; ++r) {
+        uint64_t idx = tbase + (uint64_t)w * (64 * IPT) + (uint64_t)r * 64 + lane;
         bool valid = idx < n;
         uint64_t k = 0, v = 0;
         uint32_t d = 0;
@@ -350,22 +354,46 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
             uint64_t vote = __ballot(valid && ((d >> b) & 1));
             m &= ((d >> b) & 1) ? vote : ~vote;
         }
-        uint64_t lower = ((uint64_t)1 << lane) - 1;
-        uint32_t wrank = (uint32_t)__popcll(m & lower);
-        bool leader = valid && ((m & lower) == 0);
-        if (leader) cnt[w * 256 + d] = (uint32_t)__popcll(m);
-        __syncthreads();
-        if (valid) {
-            uint32_t prefix = 0;
-            for (int i = 0; i < w; ++i) prefix += cnt[i * 256 + d];
-            uint32_t pos = hist[d] + run[d] + prefix + wrank;
-            sk[pos] = k;
-            if (HAS_VALS) sv[pos] = v;
-        }
-        __syncthreads();
-        run[t] += cnt[0 * 256 + t] + cnt[1 * 256 + t] + cnt[2 * 256 + t] + cnt[3 * 256 + t];
-        __syncthreads();
+        int leader_lane = (int)__ffsll((unsigned long long)m) - 1;
+        if (leader_lane < 0) leader_lane = 0;
+        uint32_t base = 0;
+        if (valid && lane == leader_lane)
+            base = atomicAdd(&whist[w * 256 + d], (uint32_t)__popcll(m));
+        base = __shfl(base, leader_lane);
+        kk[r] = k;
+        if (HAS_VALS) vv[r] = v;
+        dd[r] = (uint16_t)d;
+        rank[r] = base + (uint32_t)__popcll(m & lower);
     }
+    __syncthreads();
+
+    /* cross-wave digit offsets: whist[w][d] <- sum of waves < w (in place) */
+    {
+        uint32_t c0 = whist[t], c1 = whist[256 + t], c2 = whist[512 + t];
+        __syncthreads();
+        whist[t] = 0;
+        whist[256 + t] = c0;
+        whist[512 + t] = c0 + c1;
+        whist[768 + t] = c0 + c1 + c2;
+    }
+    __syncthreads();
+
+    /* reorder into LDS at the stable tile-local position */
+    {
+        uint32_t chunk0 = (uint32_t)w * (64 * IPT);
+        uint32_t chunk_n = tile_n > chunk0 ? tile_n - chunk0 : 0;
+#pragma unroll
+        for (int r = 0; r < IPT; ++r) {
+            uint32_t local = (uint32_t)r * 64 + lane;
+            if (local < chunk_n) {
+                uint32_t d = dd[r];
+                uint32_t pos = hist[d] + whist[w * 256 + d] + rank[r];
+                sk[pos] = kk[r];
+                if (HAS_VALS) sv[pos] = vv[r];
+            }
+        }
+    }
+    __syncthreads();
 
     /* write out: LDS-linear read -> digit-contiguous global writes */
     for (uint32_t p = t; p < tile_n; p += BLOCK) {
@@ -383,25 +411,27 @@ static hipError_t scatter_pass(hipStream_t s, const uint64_t *in_k, const uint64
                                bool has_vals, uint32_t ndigits, DF df, Ws &ws,
                                const char *prof_name) {
     uint32_t nb = nblocks_for(n);
+    Ws w2 = ws; /* transient: raw histogram + scan scratch, reused per pass */
+    uint32_t *raw = (uint32_t *)w2.take((size_t)nb * 256 * 4);
+    if (!raw) return hipErrorOutOfMemory;
     {
         ProfScope ps("hist", s);
-        hipLaunchKernelGGL(k_block_hist<DF>, dim3(nb), dim3(BLOCK), 0, s, in_k, n, nb, bh, df);
+        hipLaunchKernelGGL(k_block_hist<DF>, dim3(nb), dim3(BLOCK), 0, s, in_k, n, nb, bh, raw, df);
         HIP_TRY(hipGetLastError());
     }
     {
         ProfScope ps("scan", s);
-        Ws w2 = ws; /* scan scratch is transient */
         HIP_TRY(scan_u32_excl(s, bh, (uint64_t)ndigits * nb, w2));
     }
     {
         ProfScope ps(prof_name, s);
-        size_t sh = (has_vals ? 2 : 1) * (size_t)TILE * 8 + (256 + 256 + 4 * 256 + 64) * 4;
+        size_t sh = (has_vals ? 2 : 1) * (size_t)TILE * 8 + (256 + 4 * 256 + 64) * 4;
         if (has_vals)
             hipLaunchKernelGGL((k_scatter<DF, true>), dim3(nb), dim3(BLOCK), sh, s,
-                               in_k, in_v, n, nb, bh, out_k, out_v, df);
+                               in_k, in_v, n, nb, bh, raw, out_k, out_v, df);
         else
             hipLaunchKernelGGL((k_scatter<DF, false>), dim3(nb), dim3(BLOCK), sh, s,
-                               in_k, nullptr, n, nb, bh, out_k, nullptr, df);
+                               in_k, nullptr, n, nb, bh, raw, out_k, nullptr, df);
         HIP_TRY(hipGetLastError());
     }
     return hipSuccess;
@@ -535,10 +565,15 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     uint32_t excl = inc - cnt;
     for (int i = 0; i < w; ++i) excl += wsc[i];
 
+    /* Walk the chunk. A run that both STARTS (its head) and ENDS (next head)
+     * inside this chunk is owned exclusively by this thread -> plain store
+     * (out_vv was zero/identity-initialised only for the boundary atomics).
+     * Runs crossing a chunk edge are flushed with device atomics (i64 adds
+     * wrap, so the result stays bit-exact regardless of flush order). */
     int64_t segid = (int64_t)head_base[blockIdx.x] + excl - 1;
     int64_t acc_i = (OP == 3) ? INT64_MAX : (OP == 4) ? INT64_MIN : 0;
     double acc_f = 0.0;
-    bool have = false;
+    bool have = false, started_here = false;
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
         int i = c0 + j;
@@ -548,16 +583,22 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
         uint64_t pk = (i > 0) ? sk[i - 1] : *prev;
         bool head = (gi == 0) || (key != pk);
         if (head) {
-            if (have) { /* flush previous run */
-                if (OP == 0 || OP == 1) atomicAdd((unsigned long long *)((int64_t *)out_vv + segid), (unsigned long long)acc_i);
-                else if (OP == 2) atomicAdd((double *)out_vv + segid, acc_f);
-                else if (OP == 3) atomicMin((long long *)((int64_t *)out_vv + segid), (long long)acc_i);
-                else atomicMax((long long *)((int64_t *)out_vv + segid), (long long)acc_i);
+            if (have) { /* previous run terminated by this head */
+                if (started_here) { /* exclusive: plain store */
+                    if (OP == 2) ((double *)out_vv)[segid] = acc_f;
+                    else ((int64_t *)out_vv)[segid] = acc_i;
+                } else {
+                    if (OP == 0 || OP == 1) atomicAdd((unsigned long long *)((int64_t *)out_vv + segid), (unsigned long long)acc_i);
+                    else if (OP == 2) atomicAdd((double *)out_vv + segid, acc_f);
+                    else if (OP == 3) atomicMin((long long *)((int64_t *)out_vv + segid), (long long)acc_i);
+                    else atomicMax((long long *)((int64_t *)out_vv + segid), (long long)acc_i);
+                }
             }
             segid++;
             out_k[segid] = (int64_t)key;
             acc_i = (OP == 3) ? INT64_MAX : (OP == 4) ? INT64_MIN : 0;
             acc_f = 0.0;
+            started_here = true;
         }
         have = true;
         if (OP == 0) acc_i = (int64_t)((uint64_t)acc_i + sv[i]);
@@ -566,7 +607,7 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
         else if (OP == 3) { int64_t x = (int64_t)sv[i]; acc_i = x < acc_i ? x : acc_i; }
         else { int64_t x = (int64_t)sv[i]; acc_i = x > acc_i ? x : acc_i; }
     }
-    if (have) {
+    if (have) { /* last run may continue into the next chunk: always atomic */
         if (OP == 0 || OP == 1) atomicAdd((unsigned long long *)((int64_t *)out_vv + segid), (unsigned long long)acc_i);
         else if (OP == 2) atomicAdd((double *)out_vv + segid, acc_f);
         else if (OP == 3) atomicMin((long long *)((int64_t *)out_vv + segid), (long long)acc_i);
@@ -687,6 +728,7 @@ size_t ws_bytes_for(uint64_t n) {
     size_t b = 0;
     b += 4 * ((n * 8 + 255) & ~255ULL);           /* sort ping-pong k+v */
     b += ((size_t)256 * nb * 4 + 255) & ~255ULL;  /* bh matrix */
+    b += ((size_t)256 * nb * 4 + 255) & ~255ULL;  /* raw block hists (transient) */
     b += 8 * 256 * 4 + 256;                       /* hist8 */
     b += (((size_t)nb + 2) * 4 + 255) & ~255ULL;  /* head counts */
     /* scan recursion partials: nb/TILE + nb/TILE^2 + ... < nb/2048 */
