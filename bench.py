@@ -96,6 +96,13 @@ def main():
     ap.add_argument("--key-bits", type=int, default=63)
     ap.add_argument("--seed", type=int, default=0xC0FFEE + 1)  # config C1
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--op", choices=["reduce", "group_count", "sort"],
+                    default="reduce",
+                    help="reduce = C1 (driver default); group_count = C2 "
+                         "semantics (with map-side pre-combine before the "
+                         "exchange); sort = C3 sort_by_key")
+    ap.add_argument("--dist", choices=["uniform", "zipf"], default="uniform")
+    ap.add_argument("--zipf-keyspace", type=int, default=100_000_000)
     args = ap.parse_args()
 
     import torch
@@ -117,31 +124,72 @@ def main():
     dev = torch.device("cuda")
 
     # ---- setup (untimed): generate the shard resident in HBM ----
-    k = torch.empty(rows, dtype=torch.int64, device=dev)
-    v = torch.empty(rows, dtype=torch.int64, device=dev)
-    gpu.dev_gen_uniform(k, v, seed=args.seed, key_bits=args.key_bits,
-                        start=rank * rows)
+    if args.dist == "zipf":
+        from vega_amd import datagen
+        hk, hv = datagen.zipf_pairs(args.seed, rows, s=1.1,
+                                    keyspace=args.zipf_keyspace, start=rank * rows)
+        k = torch.from_numpy(hk).to(dev)
+        v = torch.from_numpy(hv).to(dev)
+    else:
+        k = torch.empty(rows, dtype=torch.int64, device=dev)
+        v = torch.empty(rows, dtype=torch.int64, device=dev)
+        gpu.dev_gen_uniform(k, v, seed=args.seed, key_bits=args.key_bits,
+                            start=rank * rows)
     slack = 1.10 if world > 1 else 1.0
     cap = int(rows * slack) + 1024
     ws = gpu.alloc_ws(cap)
     out_k = torch.empty(cap, dtype=torch.int64, device=dev)
     out_v = torch.empty(cap, dtype=torch.int64, device=dev)
-    if world > 1:
+    if world > 1 or args.op == "sort":
         pk = torch.empty(rows, dtype=torch.int64, device=dev)
         pv = torch.empty(rows, dtype=torch.int64, device=dev)
 
     nout = None
 
-    def step():
+    def step_reduce(op):
         nonlocal nout
         if world == 1:
-            nout = gpu.dev_sort_reduce(k, v, gpu.OP_SUM_I64, out_k, out_v, ws)
+            nout = gpu.dev_sort_reduce(k, v, op, out_k, out_v, ws)
         else:
             counts = gpu.dev_partition(k, v, world, pk, pv, ws)
             rk, rv = shuffle.all_to_all_kv(pk, pv, counts.astype(np.int64).tolist())
             if rk.numel() > cap:
                 raise RuntimeError(f"recv shard {rk.numel()} > cap {cap}")
-            nout = gpu.dev_sort_reduce(rk, rv, gpu.OP_SUM_I64, out_k, out_v, ws)
+            nout = gpu.dev_sort_reduce(rk, rv, op, out_k, out_v, ws)
+
+    def step_group_count():
+        # C2 with map-side pre-combine (vega's own map-side combine,
+        # dependency.rs:191-210): local (key,count) aggregate BEFORE the
+        # exchange, so Zipf hot keys cross xGMI as one row per rank
+        nonlocal nout
+        if world == 1:
+            nout = gpu.dev_sort_reduce(k, v, gpu.OP_COUNT, out_k, out_v, ws)
+            return
+        nagg = gpu.dev_sort_reduce(k, v, gpu.OP_COUNT, out_k, out_v, ws)
+        counts = gpu.dev_partition(out_k[:nagg], out_v[:nagg], world, pk, pv, ws)
+        rk, rv = shuffle.all_to_all_kv(pk[:nagg], pv[:nagg],
+                                       counts.astype(np.int64).tolist())
+        nout = gpu.dev_sort_reduce(rk, rv, gpu.OP_SUM_I64, out_k, out_v, ws)
+
+    def step_sort():
+        nonlocal nout
+        if world == 1:
+            out_k[:rows].copy_(k)
+            out_v[:rows].copy_(v)
+            gpu.dev_sort_pairs(out_k[:rows], out_v[:rows], ws)
+            nout = rows
+        else:
+            spl = shuffle.choose_splitters(k, world)
+            counts = gpu.dev_partition_range(k, v, spl, pk, pv, ws)
+            rk, rv = shuffle.all_to_all_kv(pk, pv, counts.astype(np.int64).tolist())
+            if rk.numel() > cap:
+                raise RuntimeError(f"recv shard {rk.numel()} > cap {cap}")
+            gpu.dev_sort_pairs(rk, rv, ws)
+            nout = rk.numel()
+
+    step = {"reduce": lambda: step_reduce(gpu.OP_SUM_I64),
+            "group_count": step_group_count,
+            "sort": step_sort}[args.op]
 
     def barrier_sync():
         torch.cuda.synchronize()
@@ -212,7 +260,9 @@ def main():
             "dtype": "int64",
             "data": "synthetic",
             "config": {
-                "workload": WORKLOAD,
+                "workload": WORKLOAD if args.op == "reduce" and args.dist == "uniform"
+                else f"op={args.op} dist={args.dist} rows/GPU={rows} "
+                     f"(C2/C3-style; the driver-default line is C1)",
                 "rows_per_gpu": rows,
                 "key_bits": args.key_bits,
                 "nparts": 256,

@@ -134,6 +134,14 @@ int vega_dev_partition_i64(void *stream, const int64_t *keys, const int64_t *val
                            int64_t *out_k, int64_t *out_v,
                            uint64_t *h_counts, void *d_ws, size_t ws_bytes);
 
+/* range partition for sort_by_key's exchange (Spark-style range
+ * partitioner): bucket = # splitters <= key (signed order); d_splitters is
+ * a DEVICE array of nparts-1 ascending splitters. */
+int vega_dev_partition_range_i64(void *stream, const int64_t *keys, const int64_t *vals,
+                                 uint64_t n, uint32_t nparts, const int64_t *d_splitters,
+                                 int64_t *out_k, int64_t *out_v, uint64_t *h_counts,
+                                 void *d_ws, size_t ws_bytes);
+
 /* reduce-side sort+segmented-aggregate (replaces shuffled_rdd.rs:154-164's
  * HashMap merge_combiners): stable LSB radix sort of (k,v) by key then one
  * combiner per equal-key run. in_k/in_v are NOT modified. out arrays must

@@ -225,6 +225,28 @@ def test_dev_sort_reduce_torch(ctx):
     assert got == sorted_pairs(ok, ov)
 
 
+def test_dev_partition_range_torch(ctx):
+    import torch
+    from vega_amd import gpu, shuffle
+    n = 500_000
+    k = torch.empty(n, dtype=torch.int64, device="cuda")
+    v = torch.empty(n, dtype=torch.int64, device="cuda")
+    gpu.dev_gen_uniform(k, v, seed=171, key_bits=64)
+    spl_np = np.sort(np.random.RandomState(0).randint(-2**62, 2**62, size=7))
+    spl = torch.from_numpy(spl_np).cuda()
+    ok = torch.empty_like(k)
+    ov = torch.empty_like(v)
+    ws = gpu.alloc_ws(n)
+    counts = gpu.dev_partition_range(k, v, spl, ok, ov, ws)
+    assert counts.sum() == n
+    hk = ok.cpu().numpy()
+    # matches the CPU reference partition exactly (stable + same buckets)
+    rk, rv, rcounts = shuffle.partition_range_cpu(k.cpu().numpy(), v.cpu().numpy(), spl_np)
+    assert (counts.astype(np.int64) == rcounts).all()
+    assert (hk == rk).all()
+    assert (ov.cpu().numpy() == rv).all()
+
+
 def test_checksum_matches_oracle(ctx):
     import torch
     from vega_amd import gpu

@@ -354,6 +354,17 @@ int vega_dev_partition_i64(void *stream, const int64_t *keys, const int64_t *val
     return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
 }
 
+int vega_dev_partition_range_i64(void *stream, const int64_t *keys, const int64_t *vals,
+                                 uint64_t n, uint32_t nparts, const int64_t *d_splitters,
+                                 int64_t *out_k, int64_t *out_v, uint64_t *h_counts,
+                                 void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    hipError_t e = range_partition((hipStream_t)stream, (const uint64_t *)keys,
+                                   (const uint64_t *)vals, n, nparts, d_splitters,
+                                   (uint64_t *)out_k, (uint64_t *)out_v, h_counts, ws);
+    return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
+}
+
 int vega_dev_sort_reduce(void *stream, const int64_t *in_k, const void *in_v,
                          uint64_t n, int op, int64_t *out_k, void *out_v,
                          uint64_t *h_nout, void *d_ws, size_t ws_bytes) {

@@ -65,6 +65,11 @@ hipError_t hash_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *i
                           uint64_t n, uint32_t nparts, uint64_t *out_k, uint64_t *out_v,
                           uint64_t *h_counts, Ws &ws);
 
+/* range partition (sort exchange): bucket = #splitters <= key, signed */
+hipError_t range_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                           uint64_t n, uint32_t nparts, const int64_t *d_splitters,
+                           uint64_t *out_k, uint64_t *out_v, uint64_t *h_counts, Ws &ws);
+
 hipError_t gen_uniform(hipStream_t s, int64_t *keys, int64_t *vals, uint64_t n,
                        uint64_t seed, int key_bits, uint64_t start);
 

@@ -131,6 +131,21 @@ struct RadixDigitTopSigned {
         return ((uint32_t)(k >> 56) & 0xFFu) ^ 0x80u;
     }
 };
+/* range partition for sort_by_key's exchange: bucket = # splitters <= key
+ * (signed compare; splitters ascending, np-1 of them, tiny -> L2/L1 cached) */
+struct RangeDigit {
+    const int64_t *splitters;
+    uint32_t np;
+    __device__ uint32_t operator()(uint64_t k) const {
+        int64_t key = (int64_t)k;
+        uint32_t lo = 0, hi = np - 1;
+        while (lo < hi) {
+            uint32_t m = (lo + hi) >> 1;
+            if (splitters[m] <= key) lo = m + 1; else hi = m;
+        }
+        return lo;
+    }
+};
 
 /* ------------------------------------------------------------------ */
 /* generator / elementwise                                             */
@@ -669,9 +684,10 @@ __global__ void k_gather_starts(const uint32_t *bh_scanned, uint32_t nblocks,
     if (p < (int)nparts) starts[p] = bh_scanned[(uint64_t)p * nblocks];
 }
 
-hipError_t hash_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
-                          uint64_t n, uint32_t nparts, uint64_t *out_k, uint64_t *out_v,
-                          uint64_t *h_counts, Ws &ws) {
+template <class DF>
+static hipError_t partition_generic(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                                    uint64_t n, uint32_t nparts, uint64_t *out_k,
+                                    uint64_t *out_v, uint64_t *h_counts, DF df, Ws &ws) {
     if (nparts == 0 || nparts > 256) return hipErrorInvalidValue;
     if (n == 0) {
         for (uint32_t p = 0; p < nparts; ++p) h_counts[p] = 0;
@@ -681,7 +697,6 @@ hipError_t hash_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *bh = (uint32_t *)ws.take((size_t)nparts * nb * 4);
     uint32_t *starts = (uint32_t *)ws.take((size_t)(nparts + 1) * 4);
     if (!bh || !starts) return hipErrorOutOfMemory;
-    HashModDigit df{nparts};
     HIP_TRY(scatter_pass(s, in_k, in_v, n, bh, out_k, out_v, in_v != nullptr, nparts, df, ws,
                          "partition_scatter"));
     hipLaunchKernelGGL(k_gather_starts, dim3((nparts + 255) / 256), dim3(256), 0, s,
@@ -695,6 +710,20 @@ hipError_t hash_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         h_counts[p] = next - hs[p];
     }
     return hipSuccess;
+}
+
+hipError_t hash_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                          uint64_t n, uint32_t nparts, uint64_t *out_k, uint64_t *out_v,
+                          uint64_t *h_counts, Ws &ws) {
+    return partition_generic(s, in_k, in_v, n, nparts, out_k, out_v, h_counts,
+                             HashModDigit{nparts}, ws);
+}
+
+hipError_t range_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                           uint64_t n, uint32_t nparts, const int64_t *d_splitters,
+                           uint64_t *out_k, uint64_t *out_v, uint64_t *h_counts, Ws &ws) {
+    return partition_generic(s, in_k, in_v, n, nparts, out_k, out_v, h_counts,
+                             RangeDigit{d_splitters, nparts}, ws);
 }
 
 /* ------------------------------------------------------------------ */

@@ -236,6 +236,20 @@ def dev_partition(keys_t, vals_t, nparts, out_k, out_v, ws_t):
     return counts
 
 
+def dev_partition_range(keys_t, vals_t, splitters_t, out_k, out_v, ws_t):
+    """range partition: bucket = #splitters <= key; splitters_t is a device
+    tensor of nparts-1 ascending i64 splitters"""
+    n = keys_t.numel()
+    nparts = splitters_t.numel() + 1
+    counts = np.zeros(nparts, dtype=np.uint64)
+    _check(lib().vega_dev_partition_range_i64(
+        _stream(), _t(keys_t), _t(vals_t), ctypes.c_uint64(n),
+        ctypes.c_uint32(nparts), _t(splitters_t), _t(out_k), _t(out_v),
+        _pp(counts), _t(ws_t), ctypes.c_size_t(ws_t.numel())),
+        "dev_partition_range")
+    return counts
+
+
 def dev_sort_reduce(keys_t, vals_t, op, out_k, out_v, ws_t):
     n = keys_t.numel()
     nout = ctypes.c_uint64()

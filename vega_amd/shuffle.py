@@ -40,6 +40,41 @@ def partition_cpu(k, v, nparts):
     return k[order], v[order], counts
 
 
+def choose_splitters(keys, world, group=None, samples=4096):
+    """Range-partitioner boundaries for sort_by_key's exchange (the analogue
+    of Spark's RangePartitioner sampling; reference has only take_ordered,
+    rdd.rs:1106-1153): every rank contributes `samples` strided key samples,
+    ranks all-gather them, and the world-1 quantiles become the splitters.
+    Returns an int64 tensor of world-1 ascending splitters on keys.device."""
+    n = keys.numel()
+    if n > 0:
+        idx = torch.linspace(0, n - 1, steps=min(samples, n), dtype=torch.int64,
+                             device=keys.device)
+        local = keys[idx]
+        if local.numel() < samples:  # pad to fixed size for all_gather
+            pad = local[-1].repeat(samples - local.numel())
+            local = torch.cat([local, pad])
+    else:
+        local = torch.zeros(samples, dtype=torch.int64, device=keys.device)
+    gathered = [torch.empty_like(local) for _ in range(world)]
+    dist.all_gather(gathered, local, group=group)
+    allk = torch.cat(gathered).sort().values
+    pos = [(i * allk.numel()) // world for i in range(1, world)]
+    return allk[torch.tensor(pos, dtype=torch.int64, device=keys.device)].contiguous()
+
+
+def partition_range_cpu(k, v, splitters):
+    """CPU reference of the range partition (tests / gloo path): bucket =
+    #splitters <= key, stable reorder + counts (matches RangeDigit)."""
+    k = np.asarray(k, dtype=np.int64)
+    v = np.asarray(v, dtype=np.int64)
+    spl = np.asarray(splitters, dtype=np.int64)
+    b = np.searchsorted(spl, k, side="right")
+    order = np.argsort(b, kind="stable")
+    counts = np.bincount(b, minlength=len(spl) + 1).astype(np.int64)
+    return k[order], v[order], counts
+
+
 def all_to_all_kv(send_k, send_v, send_counts, group=None):
     """Exchange bucket-contiguous (k, v) rows: counts all-to-all, then
     payload all-to-all-v. Tensors stay on their device (CUDA -> RCCL/xGMI,
