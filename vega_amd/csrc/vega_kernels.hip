@@ -268,7 +268,7 @@ hipError_t scan_u32_excl(hipStream_t s, uint32_t *a, uint64_t n, Ws &ws) {
  * starts — saves an 8 B/row key re-read there) */
 template <class DF>
 __global__ void k_block_hist(const uint64_t *keys, uint64_t n, uint32_t nblocks,
-                             uint32_t *bh, uint32_t *raw, DF df) {
+                             uint32_t ndigits, uint32_t *bh, uint32_t *raw, DF df) {
     __shared__ uint32_t h[256];
     for (int i = threadIdx.x; i < 256; i += BLOCK) h[i] = 0;
     __syncthreads();
@@ -280,8 +280,10 @@ __global__ void k_block_hist(const uint64_t *keys, uint64_t n, uint32_t nblocks,
     }
     __syncthreads();
     for (int d = threadIdx.x; d < 256; d += BLOCK) {
-        bh[(uint64_t)d * nblocks + blockIdx.x] = h[d];
-        raw[(uint64_t)blockIdx.x * 256 + d] = h[d];
+        /* bh is carved ndigits*nblocks — writing the always-zero rows d >=
+         * ndigits would run past it (it aliased the raw area for nparts<256) */
+        if ((uint32_t)d < ndigits) bh[(uint64_t)d * nblocks + blockIdx.x] = h[d];
+        raw[(uint64_t)blockIdx.x * 256 + d] = h[d]; /* full 256: scatter scans all */
     }
 }
 
@@ -431,7 +433,8 @@ static hipError_t scatter_pass(hipStream_t s, const uint64_t *in_k, const uint64
     if (!raw) return hipErrorOutOfMemory;
     {
         ProfScope ps("hist", s);
-        hipLaunchKernelGGL(k_block_hist<DF>, dim3(nb), dim3(BLOCK), 0, s, in_k, n, nb, bh, raw, df);
+        hipLaunchKernelGGL(k_block_hist<DF>, dim3(nb), dim3(BLOCK), 0, s, in_k, n, nb,
+                           ndigits, bh, raw, df);
         HIP_TRY(hipGetLastError());
     }
     {
