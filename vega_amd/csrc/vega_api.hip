@@ -173,12 +173,12 @@ static int reduce_common(vega_ctx *c, vega_rdd_t rdd, int op, uint32_t nparts,
     if (rc) return rc;
     Ws ws(c->ws, c->ws_bytes);
     const uint64_t *sk, *sv;
-    CTX_TRY(c, radix_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
-                              r->n, true, false, ws, &sk, &sv));
+    CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
+                              r->n, ws, &sk, &sv));
     uint64_t nout = 0;
     CTX_TRY(c, seg_reduce(c->stream, sk, sv, r->n, op, (uint64_t *)o->d_k, o->d_v, &nout, ws));
     o->n = nout;
-    o->sorted = true; /* seg output is key-sorted (unsigned); collect compares sorted */
+    o->sorted = false; /* grouped (hash order), not key-sorted; collect compares sorted */
     return VEGA_OK;
 }
 
@@ -370,8 +370,8 @@ int vega_dev_sort_reduce(void *stream, const int64_t *in_k, const void *in_v,
                          uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
     Ws ws(d_ws, ws_bytes);
     const uint64_t *sk, *sv;
-    hipError_t e = radix_sort_u64((hipStream_t)stream, (const uint64_t *)in_k,
-                                  (const uint64_t *)in_v, n, true, false, ws, &sk, &sv);
+    hipError_t e = group_sort_u64((hipStream_t)stream, (const uint64_t *)in_k,
+                                  (const uint64_t *)in_v, n, ws, &sk, &sv);
     if (e != hipSuccess) return e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP;
     e = seg_reduce((hipStream_t)stream, sk, sv, n, op, (uint64_t *)out_k, out_v, h_nout, ws);
     return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);

@@ -55,6 +55,12 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
                           uint64_t n, bool has_vals, bool signed_order, Ws &ws,
                           const uint64_t **res_k, const uint64_t **res_v);
 
+/* group equal keys adjacently for the reduce path (no total-order contract):
+ * adaptive skipped key sort vs 40-bit hash sort + collision-run cleanup */
+hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                          uint64_t n, Ws &ws,
+                          const uint64_t **res_k, const uint64_t **res_v);
+
 /* segmented aggregate over key-sorted rows: one output row per equal-key
  * run. op: VEGA_OP_*. Returns #segments in *h_nout (after stream sync). */
 hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t n,

@@ -131,6 +131,16 @@ struct RadixDigitTopSigned {
         return ((uint32_t)(k >> 56) & 0xFFu) ^ 0x80u;
     }
 };
+/* digit = one byte of splitmix64(key): the GROUPING sort for reduce_by_key.
+ * Grouping needs equal keys adjacent, not a total key order, so 40 hash bits
+ * (5 passes) + a local cleanup of hash-colliding runs beat the 8-pass full
+ * key sort. */
+struct HashByteDigit {
+    int shift;
+    __device__ uint32_t operator()(uint64_t k) const {
+        return (uint32_t)(vega_hash_u64(k) >> shift) & 0xFFu;
+    }
+};
 /* range partition for sort_by_key's exchange: bucket = # splitters <= key
  * (signed compare; splitters ascending, np-1 of them, tiny -> L2/L1 cached) */
 struct RangeDigit {
@@ -509,6 +519,142 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 }
 
 /* ------------------------------------------------------------------ */
+/* grouping sort for the reduce path: adaptive key-sort vs hash40 sort  */
+
+#define VEGA_H40_MASK 0xFFFFFFFFFFULL
+
+/* local cleanup after the 5-pass hash40 sort: within each equal-h40 run,
+ * group equal keys (stable insertion sort by key). Runs are tiny (expected
+ * length 1 + n/2^40); a dirty run longer than 64 sets *err and the caller
+ * falls back to the full key sort. Cross-run reads are safe: permutations
+ * stay within a run, so every observed key keeps its run's h40. */
+__global__ void k_group_cleanup(uint64_t *k, uint64_t *v, uint64_t n,
+                                int *err) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t ki = k[i];
+        uint64_t hi_ = vega_hash_u64(ki) & VEGA_H40_MASK;
+        if (i > 0 && (vega_hash_u64(k[i - 1]) & VEGA_H40_MASK) == hi_)
+            continue; /* not a run start */
+        uint64_t j = i + 1;
+        bool dirty = false;
+        while (j < n) {
+            uint64_t kj = k[j];
+            if ((vega_hash_u64(kj) & VEGA_H40_MASK) != hi_) break;
+            dirty |= (kj != ki);
+            j++;
+        }
+        if (!dirty) continue;
+        if (j - i > 64) { *err = 1; continue; }
+        for (uint64_t x = i + 1; x < j; x++) {
+            uint64_t kx = k[x], vx = v[x];
+            uint64_t y = x;
+            while (y > i && k[y - 1] > kx) {
+                k[y] = k[y - 1];
+                v[y] = v[y - 1];
+                y--;
+            }
+            k[y] = kx;
+            v[y] = vx;
+        }
+    }
+}
+
+/* group equal keys adjacently (reduce path — no total order contract).
+ * Adaptive: if the per-byte key histograms show <= 5 active radix passes,
+ * the plain key sort is cheaper; else 5 hash-byte passes + cleanup (with a
+ * full-key-sort fallback on oversized dirty runs). */
+hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                          uint64_t n, Ws &ws,
+                          const uint64_t **res_k, const uint64_t **res_v) {
+    *res_k = in_k;
+    *res_v = in_v;
+    if (n <= 1) return hipSuccess;
+    uint32_t nb = nblocks_for(n);
+
+    uint64_t *ak = (uint64_t *)ws.take(n * 8);
+    uint64_t *bk = (uint64_t *)ws.take(n * 8);
+    uint64_t *av = (uint64_t *)ws.take(n * 8);
+    uint64_t *bv = (uint64_t *)ws.take(n * 8);
+    uint32_t *bh = (uint32_t *)ws.take((size_t)256 * nb * 4);
+    uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
+    int *d_err = (int *)ws.take(256);
+    if (!ak || !bk || !av || !bv || !bh || !h8 || !d_err) return hipErrorOutOfMemory;
+
+    HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
+    {
+        ProfScope ps("hist8", s);
+        uint32_t gb = nb < 2048 ? nb : 2048;
+        hipLaunchKernelGGL(k_hist8, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+        HIP_TRY(hipGetLastError());
+    }
+    uint32_t hh[8 * 256];
+    HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    int active = 0;
+    bool pass_on[8];
+    for (int p = 0; p < 8; ++p) {
+        int nz = 0;
+        for (int d = 0; d < 256; ++d) nz += hh[p * 256 + d] != 0;
+        pass_on[p] = nz > 1;
+        active += pass_on[p];
+    }
+
+    const uint64_t *cur_k = in_k, *cur_v = in_v;
+    if (active <= 5) { /* narrow keys: plain skipped key sort groups exactly */
+        for (int p = 0; p < 8; ++p) {
+            if (!pass_on[p]) continue;
+            uint64_t *dk = (cur_k == ak) ? bk : ak;
+            uint64_t *dv = (cur_v == av) ? bv : av;
+            RadixDigit df{8 * p};
+            HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, true, 256, df, ws, "radix_scatter"));
+            cur_k = dk;
+            cur_v = dv;
+        }
+        *res_k = cur_k;
+        *res_v = cur_v;
+        return hipSuccess;
+    }
+
+    /* hash40 grouping: 5 hash-byte passes */
+    for (int p = 0; p < 5; ++p) {
+        uint64_t *dk = (cur_k == ak) ? bk : ak;
+        uint64_t *dv = (cur_v == av) ? bv : av;
+        HashByteDigit df{8 * p};
+        HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, true, 256, df, ws, "radix_scatter"));
+        cur_k = dk;
+        cur_v = dv;
+    }
+    HIP_TRY(hipMemsetAsync(d_err, 0, 4, s));
+    {
+        ProfScope ps("group_cleanup", s);
+        uint32_t gb = nb < 2048 ? nb : 2048;
+        hipLaunchKernelGGL(k_group_cleanup, dim3(gb), dim3(BLOCK), 0, s,
+                           (uint64_t *)cur_k, (uint64_t *)cur_v, n, d_err);
+        HIP_TRY(hipGetLastError());
+    }
+    int err = 0;
+    HIP_TRY(hipMemcpyAsync(&err, d_err, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (err) { /* an oversized hash-dirty run: full 8-pass key sort instead */
+        cur_k = in_k;
+        cur_v = in_v;
+        for (int p = 0; p < 8; ++p) {
+            if (!pass_on[p]) continue;
+            uint64_t *dk = (cur_k == ak) ? bk : ak;
+            uint64_t *dv = (cur_v == av) ? bv : av;
+            RadixDigit df{8 * p};
+            HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, true, 256, df, ws, "radix_scatter"));
+            cur_k = dk;
+            cur_v = dv;
+        }
+    }
+    *res_k = cur_k;
+    *res_v = cur_v;
+    return hipSuccess;
+}
+
+/* ------------------------------------------------------------------ */
 /* segmented reduce over key-sorted rows                               */
 
 __global__ void k_head_count(const uint64_t *k, uint64_t n, uint32_t *hc) {
@@ -531,45 +677,41 @@ __global__ void k_head_count(const uint64_t *k, uint64_t n, uint32_t *hc) {
     }
 }
 
-/* OP: 0 SUM_I64, 1 COUNT, 2 SUM_F64, 3 MIN_I64, 4 MAX_I64 */
+/* OP: 0 SUM_I64, 1 COUNT, 2 SUM_F64, 3 MIN_I64, 4 MAX_I64.
+ * Register-only (no LDS staging): each thread owns one 16-row chunk; the
+ * unrolled sequential loads vectorize and every 64 B line is fully consumed
+ * by exactly one lane, so the pattern is bandwidth-clean and occupancy is
+ * not LDS-bound. */
 template <int OP>
 __global__ __launch_bounds__(BLOCK) void k_seg_emit(
-    const uint64_t *k, const void *vv, uint64_t n, const uint32_t *head_base,
-    int64_t *out_k, void *out_vv) {
-    extern __shared__ __attribute__((aligned(16))) char smem[];
-    uint64_t *sk = (uint64_t *)smem;
-    constexpr bool F64 = (OP == 2);
+    const uint64_t *__restrict__ k, const void *__restrict__ vv, uint64_t n,
+    const uint32_t *__restrict__ head_base, int64_t *__restrict__ out_k,
+    void *__restrict__ out_vv) {
+    __shared__ uint32_t wsc[BLOCK / 64];
     constexpr bool NEED_V = (OP != 1);
-    uint64_t *sv = NEED_V ? sk + TILE : nullptr;
-    uint32_t *wsc = (uint32_t *)(smem + (NEED_V ? 2 : 1) * (size_t)TILE * 8);
-    uint64_t *prev = (uint64_t *)(wsc + 64);
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t tbase = (uint64_t)blockIdx.x * TILE;
     const uint64_t *v = (const uint64_t *)vv;
+    const uint64_t c0g = tbase + (uint64_t)t * IPT; /* chunk start, global */
 
-    /* stage tile (coalesced) */
+    uint64_t kk[IPT], sv[IPT];
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
-        uint64_t idx = tbase + (uint64_t)j * BLOCK + t;
-        if (idx < n) {
-            sk[j * BLOCK + t] = k[idx];
-            if (NEED_V) sv[j * BLOCK + t] = v[idx];
-        }
+        uint64_t gi = c0g + j;
+        kk[j] = (gi < n) ? k[gi] : ~0ULL;
+        if (NEED_V) sv[j] = (gi < n) ? v[gi] : 0;
     }
-    if (t == 0) *prev = (tbase > 0) ? k[tbase - 1] : 0;
-    __syncthreads();
+    uint64_t prev = (c0g > 0 && c0g <= n) ? k[c0g - 1] : 0;
 
-    /* per-thread chunk [t*IPT, t*IPT+IPT): count heads */
-    const int c0 = t * IPT;
+    /* count heads in the chunk */
     uint32_t cnt = 0;
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
-        int i = c0 + j;
-        uint64_t gi = tbase + i;
+        uint64_t gi = c0g + j;
         if (gi < n) {
-            uint64_t pk = (i > 0) ? sk[i - 1] : *prev;
-            cnt += (gi == 0) || (sk[i] != pk);
+            uint64_t pk = (j > 0) ? kk[j - 1] : prev;
+            cnt += (gi == 0) || (kk[j] != pk);
         }
     }
     /* block exclusive scan of cnt */
@@ -594,11 +736,10 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     bool have = false, started_here = false;
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
-        int i = c0 + j;
-        uint64_t gi = tbase + i;
+        uint64_t gi = c0g + j;
         if (gi >= n) break;
-        uint64_t key = sk[i];
-        uint64_t pk = (i > 0) ? sk[i - 1] : *prev;
+        uint64_t key = kk[j];
+        uint64_t pk = (j > 0) ? kk[j - 1] : prev;
         bool head = (gi == 0) || (key != pk);
         if (head) {
             if (have) { /* previous run terminated by this head */
@@ -619,11 +760,11 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
             started_here = true;
         }
         have = true;
-        if (OP == 0) acc_i = (int64_t)((uint64_t)acc_i + sv[i]);
+        if (OP == 0) acc_i = (int64_t)((uint64_t)acc_i + sv[j]);
         else if (OP == 1) acc_i += 1;
-        else if (OP == 2) acc_f += __longlong_as_double((long long)sv[i]);
-        else if (OP == 3) { int64_t x = (int64_t)sv[i]; acc_i = x < acc_i ? x : acc_i; }
-        else { int64_t x = (int64_t)sv[i]; acc_i = x > acc_i ? x : acc_i; }
+        else if (OP == 2) acc_f += __longlong_as_double((long long)sv[j]);
+        else if (OP == 3) { int64_t x = (int64_t)sv[j]; acc_i = x < acc_i ? x : acc_i; }
+        else { int64_t x = (int64_t)sv[j]; acc_i = x > acc_i ? x : acc_i; }
     }
     if (have) { /* last run may continue into the next chunk: always atomic */
         if (OP == 0 || OP == 1) atomicAdd((unsigned long long *)((int64_t *)out_vv + segid), (unsigned long long)acc_i);
@@ -663,7 +804,7 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     }
     {
         ProfScope ps("seg_emit", s);
-        size_t sh = (op == 1 ? 1 : 2) * (size_t)TILE * 8 + 64 * 4 + 16;
+        size_t sh = 0;
         switch (op) {
         case 0: hipLaunchKernelGGL(k_seg_emit<0>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v); break;
         case 1: hipLaunchKernelGGL(k_seg_emit<1>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v); break;
