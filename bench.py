@@ -74,13 +74,15 @@ def cpu_baseline_leg(rows, key_bits, seed):
     }
 
 
-def read_traffic_calibration():
-    """Optional: measured per-launch HBM bytes for the dominant kernel from a
-    committed rocprofv3 PMC calibration (profiles/traffic_calib.json)."""
+def read_traffic_calibration(rows):
+    """Measured HBM bytes per launch for the dominant kernel, scaled from the
+    committed rocprofv3 PMC calibration (profiles/traffic_calib.json holds
+    bytes/row measured via FETCH_SIZE[x2 gfx950 correction]+WRITE_SIZE)."""
     p = os.path.join(ROOT, "profiles", "traffic_calib.json")
     if os.path.exists(p):
         try:
-            return json.load(open(p)).get("radix_scatter_bytes_per_launch")
+            bpr = json.load(open(p)).get("radix_scatter_hbm_bytes_per_row")
+            return round(bpr * rows / 1e9, 2) if bpr else None  # GB per launch
         except Exception:
             return None
     return None
@@ -238,7 +240,7 @@ def main():
                 "peak": HBM_PEAK_GBS,
                 "unit": "GB/s",
                 "frac": round(achieved / HBM_PEAK_GBS, 4),
-                "traffic": read_traffic_calibration(),
+                "traffic": read_traffic_calibration(rows),
                 "kernel": "radix_scatter",
                 "launches": sc["n"],
                 "avg_launch_ms": round(avg_ms, 3),

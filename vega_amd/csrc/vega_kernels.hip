@@ -696,11 +696,27 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     const uint64_t c0g = tbase + (uint64_t)t * IPT; /* chunk start, global */
 
     uint64_t kk[IPT], sv[IPT];
+    if (c0g + IPT <= n) { /* interior chunk: unguarded 16-B vector loads */
+        const ulonglong2 *kp = reinterpret_cast<const ulonglong2 *>(k + c0g);
+        const ulonglong2 *vp = reinterpret_cast<const ulonglong2 *>(v + c0g);
 #pragma unroll
-    for (int j = 0; j < IPT; ++j) {
-        uint64_t gi = c0g + j;
-        kk[j] = (gi < n) ? k[gi] : ~0ULL;
-        if (NEED_V) sv[j] = (gi < n) ? v[gi] : 0;
+        for (int j2 = 0; j2 < IPT / 2; ++j2) {
+            ulonglong2 t2 = kp[j2];
+            kk[2 * j2] = t2.x;
+            kk[2 * j2 + 1] = t2.y;
+            if (NEED_V) {
+                ulonglong2 u2 = vp[j2];
+                sv[2 * j2] = u2.x;
+                sv[2 * j2 + 1] = u2.y;
+            }
+        }
+    } else {
+#pragma unroll
+        for (int j = 0; j < IPT; ++j) {
+            uint64_t gi = c0g + j;
+            kk[j] = (gi < n) ? k[gi] : ~0ULL;
+            if (NEED_V) sv[j] = (gi < n) ? v[gi] : 0;
+        }
     }
     uint64_t prev = (c0g > 0 && c0g <= n) ? k[c0g - 1] : 0;
 
@@ -766,7 +782,29 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
         else if (OP == 3) { int64_t x = (int64_t)sv[j]; acc_i = x < acc_i ? x : acc_i; }
         else { int64_t x = (int64_t)sv[j]; acc_i = x > acc_i ? x : acc_i; }
     }
-    if (have) { /* last run may continue into the next chunk: always atomic */
+    /* Tail flush. A long (hot-key) run makes MANY consecutive chunks
+     * head-free, all flushing into the same segid — combine those within
+     * the wave first (segmented inclusive shfl-scan over the ascending,
+     * contiguous equal-segid lanes; one atomic per group instead of 64). */
+    long long sid = (have && cnt == 0) ? (long long)segid : -(long long)(lane + 2);
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+        long long s2 = __shfl_up(sid, off);
+        if (OP == 2) {
+            double a2 = __shfl_up(acc_f, off);
+            if (lane >= off && s2 == sid) acc_f += a2;
+        } else {
+            long long a2 = __shfl_up((long long)acc_i, off);
+            if (lane >= off && s2 == sid) {
+                if (OP == 3) acc_i = (int64_t)a2 < acc_i ? (int64_t)a2 : acc_i;
+                else if (OP == 4) acc_i = (int64_t)a2 > acc_i ? (int64_t)a2 : acc_i;
+                else acc_i = (int64_t)((uint64_t)acc_i + (uint64_t)a2);
+            }
+        }
+    }
+    long long snext = __shfl_down(sid, 1);
+    bool lastg = (lane == 63) || (snext != sid);
+    if (have && (cnt != 0 || lastg)) {
         if (OP == 0 || OP == 1) atomicAdd((unsigned long long *)((int64_t *)out_vv + segid), (unsigned long long)acc_i);
         else if (OP == 2) atomicAdd((double *)out_vv + segid, acc_f);
         else if (OP == 3) atomicMin((long long *)((int64_t *)out_vv + segid), (long long)acc_i);

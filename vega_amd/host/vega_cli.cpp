@@ -39,6 +39,7 @@ int main(int argc, char **argv) {
         fprintf(stderr, "usage: vega_cli selftest\n");
         return 2;
     }
+    setvbuf(stdout, nullptr, _IONBF, 0); /* keep progress visible on a crash */
     try {
         vega::Context sc;
 
@@ -92,6 +93,15 @@ int main(int argc, char **argv) {
             pairs_t exp;
             for (auto &kv : ref) exp.push_back({kv.first, (int64_t)kv.second});
             CHECK_EQ(got, exp, "randomized reduce vs std::map (n=250k)");
+        }
+        /* join golden (test_pair_rdd.rs:40-82; col2.join(col1)) */
+        {
+            pairs_t c1{{1, 12}, {2, 34}, {3, 56}, {4, 78}};
+            pairs_t c2{{1, 101}, {1, 102}, {2, 201}, {2, 202}, {3, 301}, {3, 302}};
+            auto a = sc.parallelize(c2, 4);
+            auto b = sc.parallelize(c1, 4);
+            auto j = a.join(b, 4);
+            CHECK_EQ(j.count(), (uint64_t)6, "join golden count (6 tuples)");
         }
         /* sort_by_key: signed order + stability surrogate (sortedness) */
         {
