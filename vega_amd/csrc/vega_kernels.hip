@@ -330,12 +330,16 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *bh_scanned, const uint32_t *raw_hist,
     uint64_t *out_k, uint64_t *out_v, DF df) {
+    /* LDS stages the reordered KEYS plus a u16 source index per slot; values
+     * are gathered from the input tile through L1/L2 at writeout instead of
+     * being staged (32 KB less LDS -> 3 blocks/CU instead of 2, and the
+     * ranking loop keeps no value registers). */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *sk = (uint64_t *)smem;                              /* TILE u64 */
-    uint64_t *sv = HAS_VALS ? sk + TILE : nullptr;                /* TILE u64 */
-    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 2 : 1) * TILE * 8); /* 256 local starts */
+    uint32_t *hist = (uint32_t *)(smem + TILE * 8);               /* 256 local starts */
     uint32_t *whist = hist + 256;                                 /* 4*256 per-wave counters */
-    uint32_t *wsc = whist + 4 * 256;                              /* 4 */
+    uint32_t *wsc = whist + 4 * 256;                              /* 4 (+pad) */
+    uint16_t *sidx = (uint16_t *)(wsc + 64);                      /* TILE u16 */
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t tbase = (uint64_t)blockIdx.x * TILE;
@@ -361,18 +365,17 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
      * whist row, and hist[] is re-read only after the next barrier */
 
     /* ranking: wave w ranks rows [w*1024, w*1024+1024) in 16 rounds */
-    uint64_t kk[IPT], vv[IPT];
+    uint64_t kk[IPT];
     uint32_t rank[IPT];
     uint16_t dd[IPT];
 #pragma unroll
     for (int r = 0; r < IPT; ++r) {
         uint64_t idx = tbase + (uint64_t)w * (64 * IPT) + (uint64_t)r * 64 + lane;
         bool valid = idx < n;
-        uint64_t k = 0, v = 0;
+        uint64_t k = 0;
         uint32_t d = 0;
         if (valid) {
             k = in_k[idx];
-            if (HAS_VALS) v = in_v[idx];
             d = df(k);
         }
         uint64_t m = __ballot(valid);
@@ -388,7 +391,6 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
             base = atomicAdd(&whist[w * 256 + d], (uint32_t)__popcll(m));
         base = __shfl(base, leader_lane);
         kk[r] = k;
-        if (HAS_VALS) vv[r] = v;
         dd[r] = (uint16_t)d;
         rank[r] = base + (uint32_t)__popcll(m & lower);
     }
@@ -416,19 +418,20 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
                 uint32_t d = dd[r];
                 uint32_t pos = hist[d] + whist[w * 256 + d] + rank[r];
                 sk[pos] = kk[r];
-                if (HAS_VALS) sv[pos] = vv[r];
+                if (HAS_VALS) sidx[pos] = (uint16_t)(chunk0 + local);
             }
         }
     }
     __syncthreads();
 
-    /* write out: LDS-linear read -> digit-contiguous global writes */
+    /* write out: LDS-linear reads -> digit-contiguous global writes; values
+     * gathered from the (L1/L2-hot) input tile via the staged source index */
     for (uint32_t p = t; p < tile_n; p += BLOCK) {
         uint64_t k = sk[p];
         uint32_t d = df(k);
         uint64_t gpos = (uint64_t)bh_scanned[(uint64_t)d * nblocks + blockIdx.x] + (p - hist[d]);
         out_k[gpos] = k;
-        if (HAS_VALS) out_v[gpos] = sv[p];
+        if (HAS_VALS) out_v[gpos] = in_v[tbase + sidx[p]];
     }
 }
 
@@ -453,7 +456,7 @@ static hipError_t scatter_pass(hipStream_t s, const uint64_t *in_k, const uint64
     }
     {
         ProfScope ps(prof_name, s);
-        size_t sh = (has_vals ? 2 : 1) * (size_t)TILE * 8 + (256 + 4 * 256 + 64) * 4;
+        size_t sh = (size_t)TILE * 8 + (256 + 4 * 256 + 64) * 4 + (size_t)TILE * 2;
         if (has_vals)
             hipLaunchKernelGGL((k_scatter<DF, true>), dim3(nb), dim3(BLOCK), sh, s,
                                in_k, in_v, n, nb, bh, raw, out_k, out_v, df);
