@@ -157,6 +157,20 @@ struct RangeDigit {
     }
 };
 
+/* lanes of this wave holding the same 8-bit digit (valid lanes only).
+ * Dedupes histogram atomics: the leader adds popcount(mask) once, so a
+ * skewed (hot-key) digit costs one LDS atomic per wave, not 64 serialized
+ * same-address atomics. */
+__device__ __forceinline__ uint64_t wave_match8(uint32_t d, bool valid) {
+    uint64_t m = __ballot(valid);
+#pragma unroll
+    for (int b = 0; b < 8; ++b) {
+        uint64_t vote = __ballot(valid && ((d >> b) & 1));
+        m &= ((d >> b) & 1) ? vote : ~vote;
+    }
+    return m;
+}
+
 /* ------------------------------------------------------------------ */
 /* generator / elementwise                                             */
 
@@ -282,11 +296,16 @@ __global__ void k_block_hist(const uint64_t *keys, uint64_t n, uint32_t nblocks,
     __shared__ uint32_t h[256];
     for (int i = threadIdx.x; i < 256; i += BLOCK) h[i] = 0;
     __syncthreads();
+    const int lane = threadIdx.x & 63;
+    const uint64_t lower = ((uint64_t)1 << lane) - 1;
     uint64_t tbase = (uint64_t)blockIdx.x * TILE;
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
         uint64_t idx = tbase + (uint64_t)j * BLOCK + threadIdx.x;
-        if (idx < n) atomicAdd(&h[df(keys[idx])], 1u);
+        bool valid = idx < n;
+        uint32_t d = valid ? df(keys[idx]) : 0;
+        uint64_t m = wave_match8(d, valid);
+        if (valid && (m & lower) == 0) atomicAdd(&h[d], (uint32_t)__popcll(m));
     }
     __syncthreads();
     for (int d = threadIdx.x; d < 256; d += BLOCK) {
@@ -302,11 +321,19 @@ __global__ void k_hist8(const uint64_t *keys, uint64_t n, uint32_t *h8) {
     __shared__ uint32_t h[8][256];
     for (int i = threadIdx.x; i < 8 * 256; i += BLOCK) ((uint32_t *)h)[i] = 0;
     __syncthreads();
+    const int lane = threadIdx.x & 63;
+    const uint64_t lower = ((uint64_t)1 << lane) - 1;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        uint64_t k = keys[i];
+    uint64_t end = ((n + 63) / 64) * 64; /* whole waves, tail lanes invalid */
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < end; i += stride) {
+        bool valid = i < n;
+        uint64_t k = valid ? keys[i] : 0;
 #pragma unroll
-        for (int b = 0; b < 8; ++b) atomicAdd(&h[b][(k >> (8 * b)) & 0xFF], 1u);
+        for (int b = 0; b < 8; ++b) {
+            uint32_t d = (uint32_t)(k >> (8 * b)) & 0xFF;
+            uint64_t m = wave_match8(d, valid);
+            if (valid && (m & lower) == 0) atomicAdd(&h[b][d], (uint32_t)__popcll(m));
+        }
     }
     __syncthreads();
     for (int i = threadIdx.x; i < 8 * 256; i += BLOCK) {
@@ -378,12 +405,7 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
             k = in_k[idx];
             d = df(k);
         }
-        uint64_t m = __ballot(valid);
-#pragma unroll
-        for (int b = 0; b < 8; ++b) {
-            uint64_t vote = __ballot(valid && ((d >> b) & 1));
-            m &= ((d >> b) & 1) ? vote : ~vote;
-        }
+        uint64_t m = wave_match8(d, valid);
         int leader_lane = (int)__ffsll((unsigned long long)m) - 1;
         if (leader_lane < 0) leader_lane = 0;
         uint32_t base = 0;
