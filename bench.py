@@ -98,7 +98,7 @@ def main():
     ap.add_argument("--key-bits", type=int, default=63)
     ap.add_argument("--seed", type=int, default=0xC0FFEE + 1)  # config C1
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--op", choices=["reduce", "group_count", "sort"],
+    ap.add_argument("--op", choices=["reduce", "group_count", "sort", "join"],
                     default="reduce",
                     help="reduce = C1 (driver default); group_count = C2 "
                          "semantics (with map-side pre-combine before the "
@@ -142,7 +142,7 @@ def main():
     ws = gpu.alloc_ws(cap)
     out_k = torch.empty(cap, dtype=torch.int64, device=dev)
     out_v = torch.empty(cap, dtype=torch.int64, device=dev)
-    if world > 1 or args.op == "sort":
+    if world > 1 or args.op in ("sort", "join"):
         pk = torch.empty(rows, dtype=torch.int64, device=dev)
         pv = torch.empty(rows, dtype=torch.int64, device=dev)
 
@@ -189,9 +189,42 @@ def main():
             gpu.dev_sort_pairs(rk, rv, ws)
             nout = rk.numel()
 
+    if args.op == "join":
+        # C4: second side, keys uniform in [0, rows) both sides
+        bk_h, bv_h = None, None
+        from vega_amd import datagen as _dg
+        ak_h, av_h = _dg.uniform_range_pairs(args.seed + 3, rows, rows, start=rank * rows)
+        bk_h, bv_h = _dg.uniform_range_pairs(args.seed + 30, rows, rows, start=rank * rows)
+        k = torch.from_numpy(ak_h).to(dev); v = torch.from_numpy(av_h).to(dev)
+        kb = torch.from_numpy(bk_h).to(dev); vb = torch.from_numpy(bv_h).to(dev)
+        jk = torch.empty(cap * 3, dtype=torch.int64, device=dev)
+        jva = torch.empty(cap * 3, dtype=torch.int64, device=dev)
+        jvb = torch.empty(cap * 3, dtype=torch.int64, device=dev)
+        sb_k = torch.empty(cap, dtype=torch.int64, device=dev)
+        sb_v = torch.empty(cap, dtype=torch.int64, device=dev)
+
+    def step_join():
+        nonlocal nout
+        if world == 1:
+            out_k[:rows].copy_(k); out_v[:rows].copy_(v)
+            sb_k[:rows].copy_(kb); sb_v[:rows].copy_(vb)
+            gpu.dev_sort_pairs(out_k[:rows], out_v[:rows], ws)
+            gpu.dev_sort_pairs(sb_k[:rows], sb_v[:rows], ws)
+            nout = gpu.dev_join_sorted(out_k[:rows], out_v[:rows],
+                                       sb_k[:rows], sb_v[:rows], jk, jva, jvb, ws)
+        else:
+            ca = gpu.dev_partition(k, v, world, pk, pv, ws)
+            rak, rav = shuffle.all_to_all_kv(pk, pv, ca.astype(np.int64).tolist())
+            cb = gpu.dev_partition(kb, vb, world, pk, pv, ws)
+            rbk, rbv = shuffle.all_to_all_kv(pk, pv, cb.astype(np.int64).tolist())
+            gpu.dev_sort_pairs(rak, rav, ws)
+            gpu.dev_sort_pairs(rbk, rbv, ws)
+            nout = gpu.dev_join_sorted(rak, rav, rbk, rbv, jk, jva, jvb, ws)
+
     step = {"reduce": lambda: step_reduce(gpu.OP_SUM_I64),
             "group_count": step_group_count,
-            "sort": step_sort}[args.op]
+            "sort": step_sort,
+            "join": step_join}[args.op]
 
     def barrier_sync():
         torch.cuda.synchronize()

@@ -67,6 +67,27 @@ def _worker(rank, port, fail_q):
             ok, ov = oc.sort_by_key_i64(allk, allv)
             assert cat_k == ok.tolist(), "C3 global key order mismatch"
             assert sorted(cat_v) == sorted(ov.tolist())
+        # ---------- C4 plan: hash-partitioned inner join ----------
+        na = nb = 8000
+        ak, av = datagen.uniform_range_pairs(777, na, 4000, start=rank * na)
+        bk, bv = datagen.uniform_range_pairs(888, nb, 4000, start=rank * nb)
+        pak, pav, ac = shuffle.partition_cpu(ak, av, WS)
+        pbk, pbv, bc = shuffle.partition_cpu(bk, bv, WS)
+        rak, rav = shuffle.all_to_all_kv(torch.from_numpy(pak.copy()),
+                                         torch.from_numpy(pav.copy()), ac.tolist())
+        rbk, rbv = shuffle.all_to_all_kv(torch.from_numpy(pbk.copy()),
+                                         torch.from_numpy(pbv.copy()), bc.tolist())
+        jk, jva, jvb = oc.join_i64(rak.numpy(), rav.numpy(), rbk.numpy(), rbv.numpy(), 1, 1)
+        obj = [None] * WS
+        dist.all_gather_object(obj, (jk.tolist(), jva.tolist(), jvb.tolist()))
+        if rank == 0:
+            got = []
+            for a, b, c in obj:
+                got += list(zip(a, b, c))
+            gak, gav = datagen.uniform_range_pairs(777, na * WS, 4000)
+            gbk, gbv = datagen.uniform_range_pairs(888, nb * WS, 4000)
+            assert sorted(got) == pyref.join(gak, gav, gbk, gbv), "C4 plan mismatch"
+
         dist.barrier()
         dist.destroy_process_group()
     except Exception as e:

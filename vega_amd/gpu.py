@@ -269,6 +269,18 @@ def dev_sort_pairs(keys_t, vals_t, ws_t):
            "dev_sort_pairs")
 
 
+def dev_join_sorted(ak, av, bk, bv, out_k, out_va, out_vb, ws_t):
+    """sort-merge inner join of two KEY-SORTED sides; returns rows emitted"""
+    nout = ctypes.c_uint64()
+    _check(lib().vega_dev_join_sorted(
+        _stream(), _t(ak), _t(av), ctypes.c_uint64(ak.numel()),
+        _t(bk), _t(bv), ctypes.c_uint64(bk.numel()),
+        _t(out_k), _t(out_va), _t(out_vb), ctypes.c_uint64(out_k.numel()),
+        ctypes.byref(nout), _t(ws_t), ctypes.c_size_t(ws_t.numel())),
+        "dev_join_sorted")
+    return nout.value
+
+
 def dev_checksum(keys_t, vals_t, ws_t):
     n = keys_t.numel()
     s = ctypes.c_uint64()
