@@ -296,16 +296,11 @@ __global__ void k_block_hist(const uint64_t *keys, uint64_t n, uint32_t nblocks,
     __shared__ uint32_t h[256];
     for (int i = threadIdx.x; i < 256; i += BLOCK) h[i] = 0;
     __syncthreads();
-    const int lane = threadIdx.x & 63;
-    const uint64_t lower = ((uint64_t)1 << lane) - 1;
     uint64_t tbase = (uint64_t)blockIdx.x * TILE;
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
         uint64_t idx = tbase + (uint64_t)j * BLOCK + threadIdx.x;
-        bool valid = idx < n;
-        uint32_t d = valid ? df(keys[idx]) : 0;
-        uint64_t m = wave_match8(d, valid);
-        if (valid && (m & lower) == 0) atomicAdd(&h[d], (uint32_t)__popcll(m));
+        if (idx < n) atomicAdd(&h[df(keys[idx])], 1u);
     }
     __syncthreads();
     for (int d = threadIdx.x; d < 256; d += BLOCK) {
@@ -316,24 +311,23 @@ __global__ void k_block_hist(const uint64_t *keys, uint64_t n, uint32_t nblocks,
     }
 }
 
+/* strided key sample for the sampled pass-planning (perf-only choice) */
+__global__ void k_sample_keys(const uint64_t *k, uint64_t n, uint32_t ns,
+                              uint64_t *out) {
+    uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < ns) out[i] = k[(uint64_t)i * (n / ns)];
+}
+
 /* all 8 byte-position histograms in one pass (radix pass skipping) */
 __global__ void k_hist8(const uint64_t *keys, uint64_t n, uint32_t *h8) {
     __shared__ uint32_t h[8][256];
     for (int i = threadIdx.x; i < 8 * 256; i += BLOCK) ((uint32_t *)h)[i] = 0;
     __syncthreads();
-    const int lane = threadIdx.x & 63;
-    const uint64_t lower = ((uint64_t)1 << lane) - 1;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    uint64_t end = ((n + 63) / 64) * 64; /* whole waves, tail lanes invalid */
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < end; i += stride) {
-        bool valid = i < n;
-        uint64_t k = valid ? keys[i] : 0;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t k = keys[i];
 #pragma unroll
-        for (int b = 0; b < 8; ++b) {
-            uint32_t d = (uint32_t)(k >> (8 * b)) & 0xFF;
-            uint64_t m = wave_match8(d, valid);
-            if (valid && (m & lower) == 0) atomicAdd(&h[b][d], (uint32_t)__popcll(m));
-        }
+        for (int b = 0; b < 8; ++b) atomicAdd(&h[b][(k >> (8 * b)) & 0xFF], 1u);
     }
     __syncthreads();
     for (int i = threadIdx.x; i < 8 * 256; i += BLOCK) {
@@ -357,16 +351,12 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *bh_scanned, const uint32_t *raw_hist,
     uint64_t *out_k, uint64_t *out_v, DF df) {
-    /* LDS stages the reordered KEYS plus a u16 source index per slot; values
-     * are gathered from the input tile through L1/L2 at writeout instead of
-     * being staged (32 KB less LDS -> 3 blocks/CU instead of 2, and the
-     * ranking loop keeps no value registers). */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *sk = (uint64_t *)smem;                              /* TILE u64 */
-    uint32_t *hist = (uint32_t *)(smem + TILE * 8);               /* 256 local starts */
+    uint64_t *sv = HAS_VALS ? sk + TILE : nullptr;                /* TILE u64 */
+    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 2 : 1) * (size_t)TILE * 8); /* 256 */
     uint32_t *whist = hist + 256;                                 /* 4*256 per-wave counters */
     uint32_t *wsc = whist + 4 * 256;                              /* 4 (+pad) */
-    uint16_t *sidx = (uint16_t *)(wsc + 64);                      /* TILE u16 */
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t tbase = (uint64_t)blockIdx.x * TILE;
@@ -392,17 +382,18 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
      * whist row, and hist[] is re-read only after the next barrier */
 
     /* ranking: wave w ranks rows [w*1024, w*1024+1024) in 16 rounds */
-    uint64_t kk[IPT];
+    uint64_t kk[IPT], vv[IPT];
     uint32_t rank[IPT];
     uint16_t dd[IPT];
 #pragma unroll
     for (int r = 0; r < IPT; ++r) {
         uint64_t idx = tbase + (uint64_t)w * (64 * IPT) + (uint64_t)r * 64 + lane;
         bool valid = idx < n;
-        uint64_t k = 0;
+        uint64_t k = 0, v = 0;
         uint32_t d = 0;
         if (valid) {
             k = in_k[idx];
+            if (HAS_VALS) v = in_v[idx];
             d = df(k);
         }
         uint64_t m = wave_match8(d, valid);
@@ -413,6 +404,7 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
             base = atomicAdd(&whist[w * 256 + d], (uint32_t)__popcll(m));
         base = __shfl(base, leader_lane);
         kk[r] = k;
+        if (HAS_VALS) vv[r] = v;
         dd[r] = (uint16_t)d;
         rank[r] = base + (uint32_t)__popcll(m & lower);
     }
@@ -440,20 +432,19 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
                 uint32_t d = dd[r];
                 uint32_t pos = hist[d] + whist[w * 256 + d] + rank[r];
                 sk[pos] = kk[r];
-                if (HAS_VALS) sidx[pos] = (uint16_t)(chunk0 + local);
+                if (HAS_VALS) sv[pos] = vv[r];
             }
         }
     }
     __syncthreads();
 
-    /* write out: LDS-linear reads -> digit-contiguous global writes; values
-     * gathered from the (L1/L2-hot) input tile via the staged source index */
+    /* write out: LDS-linear reads -> digit-contiguous global writes */
     for (uint32_t p = t; p < tile_n; p += BLOCK) {
         uint64_t k = sk[p];
         uint32_t d = df(k);
         uint64_t gpos = (uint64_t)bh_scanned[(uint64_t)d * nblocks + blockIdx.x] + (p - hist[d]);
         out_k[gpos] = k;
-        if (HAS_VALS) out_v[gpos] = in_v[tbase + sidx[p]];
+        if (HAS_VALS) out_v[gpos] = sv[p];
     }
 }
 
@@ -478,7 +469,7 @@ static hipError_t scatter_pass(hipStream_t s, const uint64_t *in_k, const uint64
     }
     {
         ProfScope ps(prof_name, s);
-        size_t sh = (size_t)TILE * 8 + (256 + 4 * 256 + 64) * 4 + (size_t)TILE * 2;
+        size_t sh = (has_vals ? 2 : 1) * (size_t)TILE * 8 + (256 + 4 * 256 + 64) * 4;
         if (has_vals)
             hipLaunchKernelGGL((k_scatter<DF, true>), dim3(nb), dim3(BLOCK), sh, s,
                                in_k, in_v, n, nb, bh, raw, out_k, out_v, df);
@@ -492,6 +483,32 @@ static hipError_t scatter_pass(hipStream_t s, const uint64_t *in_k, const uint64
 
 /* ------------------------------------------------------------------ */
 /* radix sort driver                                                   */
+
+/* estimate # of non-degenerate radix passes from a 2048-key strided sample
+ * (PERF-ONLY: callers re-verify with the exact hist8 before skipping) */
+static hipError_t sample_active_passes(hipStream_t s, const uint64_t *in_k, uint64_t n,
+                                       Ws &ws, int *a_est) {
+    uint32_t ns = (uint32_t)((n < 2048) ? n : 2048);
+    uint64_t *samp_d = (uint64_t *)ws.take(2048 * 8);
+    if (!samp_d) return hipErrorOutOfMemory;
+    hipLaunchKernelGGL(k_sample_keys, dim3((ns + 255) / 256), dim3(256), 0, s, in_k, n, ns, samp_d);
+    HIP_TRY(hipGetLastError());
+    static thread_local uint64_t samp[2048];
+    HIP_TRY(hipMemcpyAsync(samp, samp_d, (size_t)ns * 8, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    int active = 0;
+    for (int p = 0; p < 8; ++p) {
+        bool seen[256] = {false};
+        int nz = 0;
+        for (uint32_t i = 0; i < ns; ++i) {
+            uint32_t d = (uint32_t)(samp[i] >> (8 * p)) & 0xFF;
+            if (!seen[d]) { seen[d] = true; nz++; }
+        }
+        active += nz > 1;
+    }
+    *a_est = active;
+    return hipSuccess;
+}
 
 hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
                           uint64_t n, bool has_vals, bool signed_order, Ws &ws,
@@ -509,23 +526,33 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     if (!ak || !bk || !bh || !h8 || (has_vals && (!av || !bv))) return hipErrorOutOfMemory;
 
-    /* pass planning: skip byte positions where all keys share one digit */
-    HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
-    {
-        ProfScope ps("hist8", s);
-        uint32_t gb = nb < 2048 ? nb : 2048;
-        hipLaunchKernelGGL(k_hist8, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
-        HIP_TRY(hipGetLastError());
+    /* pass planning: skip byte positions where all keys share one digit.
+     * A 2048-key sample decides whether the exact hist8 can pay off at all
+     * (if the sample shows 8 active passes, nothing can be skipped). */
+    int a_est = 8;
+    HIP_TRY(sample_active_passes(s, in_k, n, ws, &a_est));
+    bool pass_on[8] = {true, true, true, true, true, true, true, true};
+    if (a_est < 8) {
+        HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
+        {
+            ProfScope ps("hist8", s);
+            uint32_t gb = nb < 2048 ? nb : 2048;
+            hipLaunchKernelGGL(k_hist8, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+            HIP_TRY(hipGetLastError());
+        }
+        uint32_t hh[8 * 256];
+        HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        for (int p = 0; p < 8; ++p) {
+            int nz = 0;
+            for (int d = 0; d < 256; ++d) nz += hh[p * 256 + d] != 0;
+            pass_on[p] = nz > 1;
+        }
     }
-    uint32_t hh[8 * 256];
-    HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
-    HIP_TRY(hipStreamSynchronize(s));
 
     const uint64_t *cur_k = in_k, *cur_v = in_v;
     for (int p = 0; p < 8; ++p) {
-        int nz = 0;
-        for (int d = 0; d < 256; ++d) nz += hh[p * 256 + d] != 0;
-        if (nz <= 1) continue; /* degenerate pass: skipping keeps stability */
+        if (!pass_on[p]) continue; /* degenerate pass: skipping keeps stability */
         uint64_t *dk = (cur_k == ak) ? bk : ak;
         uint64_t *dv = (cur_v == av) ? bv : av;
         if (p == 7 && signed_order) {
@@ -606,23 +633,31 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     int *d_err = (int *)ws.take(256);
     if (!ak || !bk || !av || !bv || !bh || !h8 || !d_err) return hipErrorOutOfMemory;
 
-    HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
-    {
-        ProfScope ps("hist8", s);
-        uint32_t gb = nb < 2048 ? nb : 2048;
-        hipLaunchKernelGGL(k_hist8, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
-        HIP_TRY(hipGetLastError());
-    }
-    uint32_t hh[8 * 256];
-    HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
-    HIP_TRY(hipStreamSynchronize(s));
-    int active = 0;
-    bool pass_on[8];
-    for (int p = 0; p < 8; ++p) {
-        int nz = 0;
-        for (int d = 0; d < 256; ++d) nz += hh[p * 256 + d] != 0;
-        pass_on[p] = nz > 1;
-        active += pass_on[p];
+    /* sampled strategy choice (perf-only; both strategies are exact): the
+     * exact hist8 is needed only when the sample suggests a skippable key
+     * sort — wide keys (C1) go straight to the hash path with no hist8 */
+    int a_est = 8;
+    HIP_TRY(sample_active_passes(s, in_k, n, ws, &a_est));
+    int active = 8;
+    bool pass_on[8] = {true, true, true, true, true, true, true, true};
+    if (a_est <= 5) {
+        HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
+        {
+            ProfScope ps("hist8", s);
+            uint32_t gb = nb < 2048 ? nb : 2048;
+            hipLaunchKernelGGL(k_hist8, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+            HIP_TRY(hipGetLastError());
+        }
+        uint32_t hh[8 * 256];
+        HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        active = 0;
+        for (int p = 0; p < 8; ++p) {
+            int nz = 0;
+            for (int d = 0; d < 256; ++d) nz += hh[p * 256 + d] != 0;
+            pass_on[p] = nz > 1;
+            active += pass_on[p];
+        }
     }
 
     const uint64_t *cur_k = in_k, *cur_v = in_v;
