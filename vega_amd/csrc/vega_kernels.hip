@@ -318,20 +318,26 @@ __global__ void k_sample_keys(const uint64_t *k, uint64_t n, uint32_t ns,
     if (i < ns) out[i] = k[(uint64_t)i * (n / ns)];
 }
 
-/* all 8 byte-position histograms in one pass (radix pass skipping) */
-__global__ void k_hist8(const uint64_t *keys, uint64_t n, uint32_t *h8) {
-    __shared__ uint32_t h[8][256];
-    for (int i = threadIdx.x; i < 8 * 256; i += BLOCK) ((uint32_t *)h)[i] = 0;
+/* all 8 byte-position histograms in one pass (radix pass skipping / the
+ * onesweep global digit bases). HASHSRC hists the splitmix64 of the key.
+ * Per-wave-private LDS copies (32 KB) cut hot-digit serialization 4x. */
+template <bool HASHSRC>
+__global__ void k_hist8t(const uint64_t *keys, uint64_t n, uint32_t *h8) {
+    __shared__ uint32_t h[4][8][256];
+    for (int i = threadIdx.x; i < 4 * 8 * 256; i += BLOCK) ((uint32_t *)h)[i] = 0;
     __syncthreads();
+    const int w = threadIdx.x >> 6;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         uint64_t k = keys[i];
+        if (HASHSRC) k = vega_hash_u64(k);
 #pragma unroll
-        for (int b = 0; b < 8; ++b) atomicAdd(&h[b][(k >> (8 * b)) & 0xFF], 1u);
+        for (int b = 0; b < 8; ++b) atomicAdd(&h[w][b][(k >> (8 * b)) & 0xFF], 1u);
     }
     __syncthreads();
+    const uint32_t *hf = (const uint32_t *)h;
     for (int i = threadIdx.x; i < 8 * 256; i += BLOCK) {
-        uint32_t v = ((uint32_t *)h)[i];
+        uint32_t v = hf[i] + hf[2048 + i] + hf[4096 + i] + hf[6144 + i];
         if (v) atomicAdd(&h8[i], v);
     }
 }
@@ -448,6 +454,168 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
     }
 }
 
+/* ------------------------------------------------------------------ */
+/* onesweep scatter: the per-pass histogram+scan kernels are replaced by a
+ * chained per-tile lookback (Merrill/Garland onesweep structure). Global
+ * per-digit bases come from ONE key/hash histogram pass; each tile's block
+ * histogram is a byproduct of ranking; the (digit, tile) prefix is resolved
+ * through a status descriptor chain. Ticket-ordered virtual tile ids make
+ * the lookback progress-safe regardless of workgroup dispatch order; the
+ * descriptor words are single 8-byte relaxed AGENT-scope atomics (sc1 — L1
+ * bypass), self-contained status+count, so no fences are needed
+ * (MI355X_MICROARCH.md §Workgroup dispatch: R2 granules). */
+
+#define OSW_ST_AGG (1ULL << 62)
+#define OSW_ST_INC (2ULL << 62)
+#define OSW_CNT_MASK ((1ULL << 62) - 1)
+
+typedef __attribute__((address_space(1))) unsigned long long gdesc_t;
+
+template <class DF, bool HAS_VALS>
+__global__ __launch_bounds__(BLOCK) void k_scatter_osw(
+    const uint64_t *in_k, const uint64_t *in_v, uint64_t n,
+    const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
+    uint64_t *out_k, uint64_t *out_v, DF df) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    uint64_t *sk = (uint64_t *)smem;                              /* TILE u64 */
+    uint64_t *sv = HAS_VALS ? sk + TILE : nullptr;                /* TILE u64 */
+    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 2 : 1) * (size_t)TILE * 8); /* 256 */
+    uint32_t *whist = hist + 256;                                 /* 4*256 */
+    uint32_t *wsc = whist + 4 * 256;                              /* 8 */
+    uint32_t *vbp = wsc + 8;                                      /* 8 */
+    uint32_t *tilebase = vbp + 8;                                 /* 256 */
+
+    const int t = threadIdx.x, lane = t & 63, w = t >> 6;
+    const uint64_t lower = ((uint64_t)1 << lane) - 1;
+
+    if (t == 0) vbp[0] = atomicAdd(ticket, 1u); /* scheduling-ordered tile id */
+    for (int i = t; i < 4 * 256; i += BLOCK) whist[i] = 0;
+    __syncthreads();
+    const uint32_t vb = vbp[0];
+    const uint64_t tbase = (uint64_t)vb * TILE;
+    const uint32_t tile_n = (uint32_t)((n - tbase < TILE) ? (n - tbase) : TILE);
+
+    /* ranking: wave w ranks rows [w*1024, w*1024+1024) in 16 rounds */
+    uint64_t kk[IPT], vv[IPT];
+    uint32_t rank[IPT];
+    uint16_t dd[IPT];
+#pragma unroll
+    for (int r = 0; r < IPT; ++r) {
+        uint64_t idx = tbase + (uint64_t)w * (64 * IPT) + (uint64_t)r * 64 + lane;
+        bool valid = idx < n;
+        uint64_t k = 0, v = 0;
+        uint32_t d = 0;
+        if (valid) {
+            k = in_k[idx];
+            if (HAS_VALS) v = in_v[idx];
+            d = df(k);
+        }
+        uint64_t m = wave_match8(d, valid);
+        int leader_lane = (int)__ffsll((unsigned long long)m) - 1;
+        if (leader_lane < 0) leader_lane = 0;
+        uint32_t base = 0;
+        if (valid && lane == leader_lane)
+            base = atomicAdd(&whist[w * 256 + d], (uint32_t)__popcll(m));
+        base = __shfl(base, leader_lane);
+        kk[r] = k;
+        if (HAS_VALS) vv[r] = v;
+        dd[r] = (uint16_t)d;
+        rank[r] = base + (uint32_t)__popcll(m & lower);
+    }
+    __syncthreads();
+
+    /* block digit counts -> publish AGGREGATE early, then local starts */
+    uint32_t c0 = whist[t], c1 = whist[256 + t], c2 = whist[512 + t], c3 = whist[768 + t];
+    uint32_t cnt = c0 + c1 + c2 + c3;
+    __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
+                       (unsigned long long)cnt | OSW_ST_AGG,
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    uint32_t inc = cnt;
+    for (int off = 1; off < 64; off <<= 1) {
+        uint32_t u = __shfl_up(inc, off);
+        if (lane >= off) inc += u;
+    }
+    if (lane == 63) wsc[w] = inc;
+    __syncthreads();
+    uint32_t excl = inc - cnt;
+    for (int i = 0; i < w; ++i) excl += wsc[i];
+    hist[t] = excl;
+    whist[t] = 0;
+    whist[256 + t] = c0;
+    whist[512 + t] = c0 + c1;
+    whist[768 + t] = c0 + c1 + c2;
+
+    /* decoupled lookback: resolve this digit's prefix over earlier tiles */
+    unsigned long long excl_tiles = 0;
+    if (vb > 0) {
+        uint64_t j = vb - 1;
+        for (;;) {
+            unsigned long long wv = __hip_atomic_load(
+                (gdesc_t *)&desc[j * 256 + t], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            unsigned long long st = wv >> 62;
+            if (st == 2) { excl_tiles += wv & OSW_CNT_MASK; break; }
+            if (st == 1) {
+                excl_tiles += wv & OSW_CNT_MASK;
+                if (j == 0) break;
+                j--;
+            } else {
+                __builtin_amdgcn_s_sleep(1);
+            }
+        }
+    }
+    __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
+                       (excl_tiles + cnt) | OSW_ST_INC,
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
+    __syncthreads();
+
+    /* reorder into LDS at the stable tile-local position */
+    {
+        uint32_t chunk0 = (uint32_t)w * (64 * IPT);
+        uint32_t chunk_n = tile_n > chunk0 ? tile_n - chunk0 : 0;
+#pragma unroll
+        for (int r = 0; r < IPT; ++r) {
+            uint32_t local = (uint32_t)r * 64 + lane;
+            if (local < chunk_n) {
+                uint32_t d = dd[r];
+                uint32_t pos = hist[d] + whist[w * 256 + d] + rank[r];
+                sk[pos] = kk[r];
+                if (HAS_VALS) sv[pos] = vv[r];
+            }
+        }
+    }
+    __syncthreads();
+
+    /* write out: digit-contiguous global writes */
+    for (uint32_t p = t; p < tile_n; p += BLOCK) {
+        uint64_t k = sk[p];
+        uint32_t d = df(k);
+        uint64_t gpos = (uint64_t)tilebase[d] + (p - hist[d]);
+        out_k[gpos] = k;
+        if (HAS_VALS) out_v[gpos] = sv[p];
+    }
+}
+
+template <class DF>
+static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                                   uint64_t n, const uint32_t *gbase_d,
+                                   unsigned long long *desc, uint32_t *ticket,
+                                   uint64_t *out_k, uint64_t *out_v, bool has_vals,
+                                   DF df, const char *prof_name) {
+    uint32_t nb = nblocks_for(n);
+    HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
+    HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
+    ProfScope ps(prof_name, s);
+    size_t sh = (has_vals ? 2 : 1) * (size_t)TILE * 8 + (256 + 4 * 256 + 8 + 8 + 256) * 4;
+    if (has_vals)
+        hipLaunchKernelGGL((k_scatter_osw<DF, true>), dim3(nb), dim3(BLOCK), sh, s,
+                           in_k, in_v, n, gbase_d, desc, ticket, out_k, out_v, df);
+    else
+        hipLaunchKernelGGL((k_scatter_osw<DF, false>), dim3(nb), dim3(BLOCK), sh, s,
+                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, df);
+    return hipGetLastError();
+}
+
 template <class DF>
 static hipError_t scatter_pass(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
                                uint64_t n, uint32_t *bh, uint64_t *out_k, uint64_t *out_v,
@@ -522,33 +690,39 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint64_t *bk = (uint64_t *)ws.take(n * 8);
     uint64_t *av = has_vals ? (uint64_t *)ws.take(n * 8) : nullptr;
     uint64_t *bv = has_vals ? (uint64_t *)ws.take(n * 8) : nullptr;
-    uint32_t *bh = (uint32_t *)ws.take((size_t)256 * nb * 4);
+    unsigned long long *desc = (unsigned long long *)ws.take((size_t)256 * nb * 8);
+    uint32_t *gbase_d = (uint32_t *)ws.take(8 * 256 * 4);
+    uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
-    if (!ak || !bk || !bh || !h8 || (has_vals && (!av || !bv))) return hipErrorOutOfMemory;
+    if (!ak || !bk || !desc || !gbase_d || !ticket || !h8 || (has_vals && (!av || !bv)))
+        return hipErrorOutOfMemory;
 
-    /* pass planning: skip byte positions where all keys share one digit.
-     * A 2048-key sample decides whether the exact hist8 can pay off at all
-     * (if the sample shows 8 active passes, nothing can be skipped). */
-    int a_est = 8;
-    HIP_TRY(sample_active_passes(s, in_k, n, ws, &a_est));
-    bool pass_on[8] = {true, true, true, true, true, true, true, true};
-    if (a_est < 8) {
-        HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
-        {
-            ProfScope ps("hist8", s);
-            uint32_t gb = nb < 2048 ? nb : 2048;
-            hipLaunchKernelGGL(k_hist8, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
-            HIP_TRY(hipGetLastError());
-        }
-        uint32_t hh[8 * 256];
-        HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
-        HIP_TRY(hipStreamSynchronize(s));
-        for (int p = 0; p < 8; ++p) {
-            int nz = 0;
-            for (int d = 0; d < 256; ++d) nz += hh[p * 256 + d] != 0;
-            pass_on[p] = nz > 1;
-        }
+    /* exact per-byte histograms: pass skipping + the onesweep global bases */
+    HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
+    {
+        ProfScope ps("hist8", s);
+        uint32_t gb = nb < 2048 ? nb : 2048;
+        hipLaunchKernelGGL(k_hist8t<false>, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+        HIP_TRY(hipGetLastError());
     }
+    uint32_t hh[8 * 256];
+    HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    bool pass_on[8];
+    static thread_local uint32_t gb_host[8 * 256];
+    for (int p = 0; p < 8; ++p) {
+        int nz = 0;
+        uint32_t running = 0;
+        for (int d = 0; d < 256; ++d) {
+            nz += hh[p * 256 + d] != 0;
+            /* base in DIGIT-FUNCTOR order: signed pass 7 emits d^0x80 */
+            int src = (p == 7 && signed_order) ? (d ^ 0x80) : d;
+            gb_host[p * 256 + d] = running;
+            running += hh[p * 256 + src];
+        }
+        pass_on[p] = nz > 1;
+    }
+    HIP_TRY(hipMemcpyAsync(gbase_d, gb_host, sizeof gb_host, hipMemcpyHostToDevice, s));
 
     const uint64_t *cur_k = in_k, *cur_v = in_v;
     for (int p = 0; p < 8; ++p) {
@@ -557,10 +731,12 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         uint64_t *dv = (cur_v == av) ? bv : av;
         if (p == 7 && signed_order) {
             RadixDigitTopSigned df{56};
-            HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, has_vals, 256, df, ws, "radix_scatter"));
+            HIP_TRY(scatter_pass_osw(s, cur_k, cur_v, n, gbase_d + p * 256, desc, ticket,
+                                     dk, dv, has_vals, df, "radix_scatter"));
         } else {
             RadixDigit df{8 * p};
-            HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, has_vals, 256, df, ws, "radix_scatter"));
+            HIP_TRY(scatter_pass_osw(s, cur_k, cur_v, n, gbase_d + p * 256, desc, ticket,
+                                     dk, dv, has_vals, df, "radix_scatter"));
         }
         cur_k = dk;
         cur_v = dv;
@@ -628,60 +804,88 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint64_t *bk = (uint64_t *)ws.take(n * 8);
     uint64_t *av = (uint64_t *)ws.take(n * 8);
     uint64_t *bv = (uint64_t *)ws.take(n * 8);
-    uint32_t *bh = (uint32_t *)ws.take((size_t)256 * nb * 4);
+    unsigned long long *desc = (unsigned long long *)ws.take((size_t)256 * nb * 8);
+    uint32_t *gbase_d = (uint32_t *)ws.take(8 * 256 * 4);
+    uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
-    if (!ak || !bk || !av || !bv || !bh || !h8 || !d_err) return hipErrorOutOfMemory;
+    if (!ak || !bk || !av || !bv || !desc || !gbase_d || !ticket || !h8 || !d_err)
+        return hipErrorOutOfMemory;
 
-    /* sampled strategy choice (perf-only; both strategies are exact): the
-     * exact hist8 is needed only when the sample suggests a skippable key
-     * sort — wide keys (C1) go straight to the hash path with no hist8 */
-    int a_est = 8;
-    HIP_TRY(sample_active_passes(s, in_k, n, ws, &a_est));
-    int active = 8;
+    static thread_local uint32_t hh[8 * 256];
+    static thread_local uint32_t gb_host[8 * 256];
     bool pass_on[8] = {true, true, true, true, true, true, true, true};
-    if (a_est <= 5) {
+
+    /* exact byte histograms of SRC(k); fills hh and gb_host (functor order
+     * == raw order here), returns #active passes */
+    auto exact_hists = [&](bool hashsrc, int *active) -> hipError_t {
         HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
         {
-            ProfScope ps("hist8", s);
+            ProfScope ps(hashsrc ? "ghist" : "hist8", s);
             uint32_t gb = nb < 2048 ? nb : 2048;
-            hipLaunchKernelGGL(k_hist8, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+            if (hashsrc)
+                hipLaunchKernelGGL(k_hist8t<true>, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+            else
+                hipLaunchKernelGGL(k_hist8t<false>, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
             HIP_TRY(hipGetLastError());
         }
-        uint32_t hh[8 * 256];
-        HIP_TRY(hipMemcpyAsync(hh, h8, sizeof hh, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipMemcpyAsync(hh, h8, 8 * 256 * 4, hipMemcpyDeviceToHost, s));
         HIP_TRY(hipStreamSynchronize(s));
-        active = 0;
+        *active = 0;
         for (int p = 0; p < 8; ++p) {
             int nz = 0;
-            for (int d = 0; d < 256; ++d) nz += hh[p * 256 + d] != 0;
+            uint32_t running = 0;
+            for (int d = 0; d < 256; ++d) {
+                nz += hh[p * 256 + d] != 0;
+                gb_host[p * 256 + d] = running;
+                running += hh[p * 256 + d];
+            }
             pass_on[p] = nz > 1;
-            active += pass_on[p];
+            *active += pass_on[p];
         }
-    }
+        return hipMemcpyAsync(gbase_d, gb_host, 8 * 256 * 4, hipMemcpyHostToDevice, s);
+    };
 
-    const uint64_t *cur_k = in_k, *cur_v = in_v;
-    if (active <= 5) { /* narrow keys: plain skipped key sort groups exactly */
+    auto run_key_passes = [&](const uint64_t **ck, const uint64_t **cv) -> hipError_t {
         for (int p = 0; p < 8; ++p) {
             if (!pass_on[p]) continue;
-            uint64_t *dk = (cur_k == ak) ? bk : ak;
-            uint64_t *dv = (cur_v == av) ? bv : av;
+            uint64_t *dk = (*ck == ak) ? bk : ak;
+            uint64_t *dv = (*cv == av) ? bv : av;
             RadixDigit df{8 * p};
-            HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, true, 256, df, ws, "radix_scatter"));
-            cur_k = dk;
-            cur_v = dv;
+            HIP_TRY(scatter_pass_osw(s, *ck, *cv, n, gbase_d + p * 256, desc, ticket,
+                                     dk, dv, true, df, "radix_scatter"));
+            *ck = dk;
+            *cv = dv;
         }
-        *res_k = cur_k;
-        *res_v = cur_v;
         return hipSuccess;
+    };
+
+    /* sampled strategy choice (perf-only; both strategies are exact) */
+    int a_est = 8;
+    HIP_TRY(sample_active_passes(s, in_k, n, ws, &a_est));
+
+    const uint64_t *cur_k = in_k, *cur_v = in_v;
+    if (a_est <= 5) {
+        int active = 8;
+        HIP_TRY(exact_hists(false, &active));
+        if (active <= 5) { /* narrow keys: skipped key sort groups exactly */
+            HIP_TRY(run_key_passes(&cur_k, &cur_v));
+            *res_k = cur_k;
+            *res_v = cur_v;
+            return hipSuccess;
+        }
+        /* sample lied; fall through to the hash path */
     }
 
     /* hash40 grouping: 5 hash-byte passes */
+    int active5 = 0;
+    HIP_TRY(exact_hists(true, &active5));
     for (int p = 0; p < 5; ++p) {
         uint64_t *dk = (cur_k == ak) ? bk : ak;
         uint64_t *dv = (cur_v == av) ? bv : av;
         HashByteDigit df{8 * p};
-        HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, true, 256, df, ws, "radix_scatter"));
+        HIP_TRY(scatter_pass_osw(s, cur_k, cur_v, n, gbase_d + p * 256, desc, ticket,
+                                 dk, dv, true, df, "radix_scatter"));
         cur_k = dk;
         cur_v = dv;
     }
@@ -696,18 +900,12 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     int err = 0;
     HIP_TRY(hipMemcpyAsync(&err, d_err, 4, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
-    if (err) { /* an oversized hash-dirty run: full 8-pass key sort instead */
+    if (err) { /* an oversized hash-dirty run: full key sort instead */
+        int active = 8;
+        HIP_TRY(exact_hists(false, &active));
         cur_k = in_k;
         cur_v = in_v;
-        for (int p = 0; p < 8; ++p) {
-            if (!pass_on[p]) continue;
-            uint64_t *dk = (cur_k == ak) ? bk : ak;
-            uint64_t *dv = (cur_v == av) ? bv : av;
-            RadixDigit df{8 * p};
-            HIP_TRY(scatter_pass(s, cur_k, cur_v, n, bh, dk, dv, true, 256, df, ws, "radix_scatter"));
-            cur_k = dk;
-            cur_v = dv;
-        }
+        HIP_TRY(run_key_passes(&cur_k, &cur_v));
     }
     *res_k = cur_k;
     *res_v = cur_v;
