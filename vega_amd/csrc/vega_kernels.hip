@@ -749,7 +749,6 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 /* ------------------------------------------------------------------ */
 /* grouping sort for the reduce path: adaptive key-sort vs hash40 sort  */
 
-#define VEGA_H40_MASK 0xFFFFFFFFFFULL
 
 /* local cleanup after the 5-pass hash40 sort: within each equal-h40 run,
  * group equal keys (stable insertion sort by key). Runs are tiny (expected
@@ -757,18 +756,18 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
  * falls back to the full key sort. Cross-run reads are safe: permutations
  * stay within a run, so every observed key keeps its run's h40. */
 __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, uint64_t n,
-                                int *err) {
+                                uint64_t hmask, int *err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         uint64_t ki = k[i];
-        uint64_t hi_ = vega_hash_u64(ki) & VEGA_H40_MASK;
-        if (i > 0 && (vega_hash_u64(k[i - 1]) & VEGA_H40_MASK) == hi_)
+        uint64_t hi_ = vega_hash_u64(ki) & hmask;
+        if (i > 0 && (vega_hash_u64(k[i - 1]) & hmask) == hi_)
             continue; /* not a run start */
         uint64_t j = i + 1;
         bool dirty = false;
         while (j < n) {
             uint64_t kj = k[j];
-            if ((vega_hash_u64(kj) & VEGA_H40_MASK) != hi_) break;
+            if ((vega_hash_u64(kj) & hmask) != hi_) break;
             dirty |= (kj != ki);
             j++;
         }
@@ -877,10 +876,14 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         /* sample lied; fall through to the hash path */
     }
 
-    /* hash40 grouping: 5 hash-byte passes */
+    /* hash grouping: enough hash bytes to keep the expected bucket load
+     * factor <= ~0.25 (4 bytes up to 2^30 rows, 5 above) — fewer passes,
+     * slightly busier cleanup */
+    const int hbytes = (n <= (1ULL << 30)) ? 4 : 5;
+    const uint64_t hmask = (hbytes >= 8) ? ~0ULL : ((1ULL << (8 * hbytes)) - 1);
     int active5 = 0;
     HIP_TRY(exact_hists(true, &active5));
-    for (int p = 0; p < 5; ++p) {
+    for (int p = 0; p < hbytes; ++p) {
         uint64_t *dk = (cur_k == ak) ? bk : ak;
         uint64_t *dv = (cur_v == av) ? bv : av;
         HashByteDigit df{8 * p};
@@ -894,7 +897,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         ProfScope ps("group_cleanup", s);
         uint32_t gb = nb < 2048 ? nb : 2048;
         hipLaunchKernelGGL(k_group_cleanup, dim3(gb), dim3(BLOCK), 0, s,
-                           (uint64_t *)cur_k, (uint64_t *)cur_v, n, d_err);
+                           (uint64_t *)cur_k, (uint64_t *)cur_v, n, hmask, d_err);
         HIP_TRY(hipGetLastError());
     }
     int err = 0;
