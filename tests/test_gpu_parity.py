@@ -112,6 +112,33 @@ def test_min_max_ops(ctx):
 
 # ---------------- group_count (C2 semantics) ----------------
 
+def test_reduce_hash_collisions(ctx):
+    """Keys whose splitmix64 hashes collide in the low 32 bits land in one
+    hash-sort run with >1 distinct key — exercises k_group_cleanup."""
+    from vega_amd import gpu, shuffle
+    cand = np.arange(1_500_000, dtype=np.int64)
+    h32 = (shuffle.hash_u64_np(cand) & np.uint64(0xFFFFFFFF)).astype(np.uint64)
+    order = np.argsort(h32, kind="stable")
+    hs = h32[order]
+    dup = hs[1:] == hs[:-1]
+    pairs = order[:-1][dup], order[1:][dup]
+    coll = np.unique(np.concatenate([cand[pairs[0]], cand[pairs[1]]]))
+    assert len(coll) >= 20, "expected some 32-bit hash collisions in 1.5M keys"
+    # each colliding key appears 3x with distinct values + filler rows
+    keys = np.concatenate([np.repeat(coll, 3),
+                           np.arange(10_000, 20_000, dtype=np.int64)])
+    vals = np.arange(len(keys), dtype=np.int64)
+    rng = np.random.RandomState(7)
+    perm = rng.permutation(len(keys))
+    keys, vals = keys[perm], vals[perm]
+    rdd = ctx.make_rdd(keys, vals)
+    red = rdd.reduce_by_key(gpu.OP_SUM_I64)
+    gk, gv = red.collect()
+    ok, ov = oc.reduce_by_key_i64(keys, vals, 8, 8)
+    assert sorted_pairs(gk, gv) == sorted_pairs(ok, ov)
+    rdd.free(); red.free()
+
+
 def test_group_count_zipf(ctx):
     n = 1_000_000
     hk, hv = datagen.zipf_pairs(21, n, s=1.1, keyspace=100_000)
