@@ -41,7 +41,7 @@
 namespace vega {
 
 constexpr int BLOCK = 256;
-constexpr int IPT = 8;             /* items per thread (TILE 2048: occupancy over run length) */
+constexpr int IPT = 16;            /* items per thread */
 constexpr int TILE = BLOCK * IPT;  /* 4096 rows per workgroup */
 
 static inline uint32_t nblocks_for(uint64_t n) {
