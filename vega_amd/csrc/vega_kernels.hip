@@ -471,19 +471,22 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
 
 typedef __attribute__((address_space(1))) unsigned long long gdesc_t;
 
-template <class DF, bool HAS_VALS>
+template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
 __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
     uint64_t *out_k, uint64_t *out_v, DF df) {
+    /* IN_PK/OUT_PK: rows are interleaved (k,v) ulonglong2 — one 16-B vector
+     * access per row instead of two 8-B streams. The first pass of a sort
+     * reads SoA and packs; the last unpacks back to SoA for free. */
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    uint64_t *sk = (uint64_t *)smem;                              /* TILE u64 */
-    uint64_t *sv = HAS_VALS ? sk + TILE : nullptr;                /* TILE u64 */
-    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 2 : 1) * (size_t)TILE * 8); /* 256 */
-    uint32_t *whist = hist + 256;                                 /* 4*256 */
-    uint32_t *wsc = whist + 4 * 256;                              /* 8 */
-    uint32_t *vbp = wsc + 8;                                      /* 8 */
-    uint32_t *tilebase = vbp + 8;                                 /* 256 */
+    ulonglong2 *spk = (ulonglong2 *)smem;                       /* TILE 16B */
+    uint64_t *sk = (uint64_t *)smem;                            /* no-vals: TILE u64 */
+    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 16 : 8) * (size_t)TILE);
+    uint32_t *whist = hist + 256;
+    uint32_t *wsc = whist + 4 * 256;
+    uint32_t *vbp = wsc + 8;
+    uint32_t *tilebase = vbp + 8;
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t lower = ((uint64_t)1 << lane) - 1;
@@ -495,21 +498,52 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
     const uint64_t tbase = (uint64_t)vb * TILE;
     const uint32_t tile_n = (uint32_t)((n - tbase < TILE) ? (n - tbase) : TILE);
 
-    /* ranking: wave w ranks rows [w*1024, w*1024+1024) in 16 rounds */
+    /* prefetch the wave's whole 1024-row chunk first (independent loads in
+     * flight together: ONE memory latency for the chunk, not one per round
+     * — the ranking rounds below are then register-only + LDS). Interior
+     * chunks take the unguarded path so nothing blocks load pipelining. */
     uint64_t kk[IPT], vv[IPT];
     uint32_t rank[IPT];
     uint16_t dd[IPT];
+    const uint64_t chunk_g = tbase + (uint64_t)w * (64 * IPT) + lane;
+    const bool chunk_full = tbase + ((uint64_t)w + 1) * (64 * IPT) <= n; /* wave-uniform */
+    if (chunk_full) {
+#pragma unroll
+        for (int r = 0; r < IPT; ++r) {
+            uint64_t idx = chunk_g + (uint64_t)r * 64;
+            if (IN_PK) {
+                ulonglong2 kv = ((const ulonglong2 *)in_k)[idx];
+                kk[r] = kv.x;
+                vv[r] = kv.y;
+            } else {
+                kk[r] = in_k[idx];
+                if (HAS_VALS) vv[r] = in_v[idx];
+            }
+        }
+    } else {
+#pragma unroll
+        for (int r = 0; r < IPT; ++r) {
+            uint64_t idx = chunk_g + (uint64_t)r * 64;
+            bool valid = idx < n;
+            kk[r] = 0;
+            vv[r] = 0;
+            if (valid) {
+                if (IN_PK) {
+                    ulonglong2 kv = ((const ulonglong2 *)in_k)[idx];
+                    kk[r] = kv.x;
+                    vv[r] = kv.y;
+                } else {
+                    kk[r] = in_k[idx];
+                    if (HAS_VALS) vv[r] = in_v[idx];
+                }
+            }
+        }
+    }
+    /* ranking: wave w ranks its chunk in 16 register-only rounds */
 #pragma unroll
     for (int r = 0; r < IPT; ++r) {
-        uint64_t idx = tbase + (uint64_t)w * (64 * IPT) + (uint64_t)r * 64 + lane;
-        bool valid = idx < n;
-        uint64_t k = 0, v = 0;
-        uint32_t d = 0;
-        if (valid) {
-            k = in_k[idx];
-            if (HAS_VALS) v = in_v[idx];
-            d = df(k);
-        }
+        bool valid = chunk_g + (uint64_t)r * 64 < n;
+        uint32_t d = valid ? df(kk[r]) : 0;
         uint64_t m = wave_match8(d, valid);
         int leader_lane = (int)__ffsll((unsigned long long)m) - 1;
         if (leader_lane < 0) leader_lane = 0;
@@ -517,8 +551,6 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
         if (valid && lane == leader_lane)
             base = atomicAdd(&whist[w * 256 + d], (uint32_t)__popcll(m));
         base = __shfl(base, leader_lane);
-        kk[r] = k;
-        if (HAS_VALS) vv[r] = v;
         dd[r] = (uint16_t)d;
         rank[r] = base + (uint32_t)__popcll(m & lower);
     }
@@ -579,8 +611,8 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
             if (local < chunk_n) {
                 uint32_t d = dd[r];
                 uint32_t pos = hist[d] + whist[w * 256 + d] + rank[r];
-                sk[pos] = kk[r];
-                if (HAS_VALS) sv[pos] = vv[r];
+                if (HAS_VALS) spk[pos] = make_ulonglong2(kk[r], vv[r]);
+                else sk[pos] = kk[r];
             }
         }
     }
@@ -588,11 +620,22 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
 
     /* write out: digit-contiguous global writes */
     for (uint32_t p = t; p < tile_n; p += BLOCK) {
-        uint64_t k = sk[p];
+        uint64_t k, v = 0;
+        if (HAS_VALS) {
+            ulonglong2 kv = spk[p];
+            k = kv.x;
+            v = kv.y;
+        } else {
+            k = sk[p];
+        }
         uint32_t d = df(k);
         uint64_t gpos = (uint64_t)tilebase[d] + (p - hist[d]);
-        out_k[gpos] = k;
-        if (HAS_VALS) out_v[gpos] = sv[p];
+        if (OUT_PK) {
+            ((ulonglong2 *)out_k)[gpos] = make_ulonglong2(k, v);
+        } else {
+            out_k[gpos] = k;
+            if (HAS_VALS) out_v[gpos] = v;
+        }
     }
 }
 
@@ -601,18 +644,29 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    uint64_t n, const uint32_t *gbase_d,
                                    unsigned long long *desc, uint32_t *ticket,
                                    uint64_t *out_k, uint64_t *out_v, bool has_vals,
+                                   bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
-    size_t sh = (has_vals ? 2 : 1) * (size_t)TILE * 8 + (256 + 4 * 256 + 8 + 8 + 256) * 4;
-    if (has_vals)
-        hipLaunchKernelGGL((k_scatter_osw<DF, true>), dim3(nb), dim3(BLOCK), sh, s,
-                           in_k, in_v, n, gbase_d, desc, ticket, out_k, out_v, df);
-    else
-        hipLaunchKernelGGL((k_scatter_osw<DF, false>), dim3(nb), dim3(BLOCK), sh, s,
+    size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 4 * 256 + 8 + 8 + 256) * 4;
+    if (!has_vals) {
+        hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(BLOCK), sh, s,
                            in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, df);
+    } else if (!in_pk && !out_pk) {
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(BLOCK), sh, s,
+                           in_k, in_v, n, gbase_d, desc, ticket, out_k, out_v, df);
+    } else if (!in_pk && out_pk) {
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(BLOCK), sh, s,
+                           in_k, in_v, n, gbase_d, desc, ticket, out_k, nullptr, df);
+    } else if (in_pk && out_pk) {
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(BLOCK), sh, s,
+                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, df);
+    } else {
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(BLOCK), sh, s,
+                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, out_v, df);
+    }
     return hipGetLastError();
 }
 
@@ -686,15 +740,13 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     if (n <= 1) return hipSuccess;
     uint32_t nb = nblocks_for(n);
 
-    uint64_t *ak = (uint64_t *)ws.take(n * 8);
-    uint64_t *bk = (uint64_t *)ws.take(n * 8);
-    uint64_t *av = has_vals ? (uint64_t *)ws.take(n * 8) : nullptr;
-    uint64_t *bv = has_vals ? (uint64_t *)ws.take(n * 8) : nullptr;
+    uint64_t *pA = (uint64_t *)ws.take(n * 16);
+    uint64_t *pB = (uint64_t *)ws.take(n * 16);
     unsigned long long *desc = (unsigned long long *)ws.take((size_t)256 * nb * 8);
     uint32_t *gbase_d = (uint32_t *)ws.take(8 * 256 * 4);
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
-    if (!ak || !bk || !desc || !gbase_d || !ticket || !h8 || (has_vals && (!av || !bv)))
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8)
         return hipErrorOutOfMemory;
 
     /* exact per-byte histograms: pass skipping + the onesweep global bases */
@@ -724,25 +776,33 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     }
     HIP_TRY(hipMemcpyAsync(gbase_d, gb_host, sizeof gb_host, hipMemcpyHostToDevice, s));
 
-    const uint64_t *cur_k = in_k, *cur_v = in_v;
-    for (int p = 0; p < 8; ++p) {
-        if (!pass_on[p]) continue; /* degenerate pass: skipping keeps stability */
-        uint64_t *dk = (cur_k == ak) ? bk : ak;
-        uint64_t *dv = (cur_v == av) ? bv : av;
+    int plist[8], np = 0;
+    for (int p = 0; p < 8; ++p)
+        if (pass_on[p]) plist[np++] = p; /* degenerate passes skipped: stability kept */
+    if (np == 0) return hipSuccess;
+    const uint64_t *cur = nullptr;
+    for (int i = 0; i < np; ++i) {
+        int p = plist[i];
+        bool in_pk = has_vals && i > 0;
+        bool out_pk = has_vals && (i < np - 1);
+        const uint64_t *src_k = (i == 0) ? in_k : cur;
+        const uint64_t *src_v = (i == 0) ? in_v : nullptr;
+        uint64_t *dbuf = ((i & 1) == 0) ? pA : pB;
+        uint64_t *dk = dbuf;
+        uint64_t *dv = (!out_pk && has_vals) ? dbuf + n : nullptr;
         if (p == 7 && signed_order) {
             RadixDigitTopSigned df{56};
-            HIP_TRY(scatter_pass_osw(s, cur_k, cur_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, has_vals, df, "radix_scatter"));
+            HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
+                                     dk, dv, has_vals, in_pk, out_pk, df, "radix_scatter"));
         } else {
             RadixDigit df{8 * p};
-            HIP_TRY(scatter_pass_osw(s, cur_k, cur_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, has_vals, df, "radix_scatter"));
+            HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
+                                     dk, dv, has_vals, in_pk, out_pk, df, "radix_scatter"));
         }
-        cur_k = dk;
-        cur_v = dv;
+        cur = dk;
     }
-    *res_k = cur_k;
-    *res_v = cur_v;
+    *res_k = cur;
+    *res_v = has_vals ? cur + n : nullptr;
     return hipSuccess;
 }
 
@@ -799,16 +859,14 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     if (n <= 1) return hipSuccess;
     uint32_t nb = nblocks_for(n);
 
-    uint64_t *ak = (uint64_t *)ws.take(n * 8);
-    uint64_t *bk = (uint64_t *)ws.take(n * 8);
-    uint64_t *av = (uint64_t *)ws.take(n * 8);
-    uint64_t *bv = (uint64_t *)ws.take(n * 8);
+    uint64_t *pA = (uint64_t *)ws.take(n * 16);
+    uint64_t *pB = (uint64_t *)ws.take(n * 16);
     unsigned long long *desc = (unsigned long long *)ws.take((size_t)256 * nb * 8);
     uint32_t *gbase_d = (uint32_t *)ws.take(8 * 256 * 4);
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
-    if (!ak || !bk || !av || !bv || !desc || !gbase_d || !ticket || !h8 || !d_err)
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err)
         return hipErrorOutOfMemory;
 
     static thread_local uint32_t hh[8 * 256];
@@ -845,17 +903,27 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         return hipMemcpyAsync(gbase_d, gb_host, 8 * 256 * 4, hipMemcpyHostToDevice, s);
     };
 
-    auto run_key_passes = [&](const uint64_t **ck, const uint64_t **cv) -> hipError_t {
-        for (int p = 0; p < 8; ++p) {
-            if (!pass_on[p]) continue;
-            uint64_t *dk = (*ck == ak) ? bk : ak;
-            uint64_t *dv = (*cv == av) ? bv : av;
+    /* run the active key passes SoA->packed->...->SoA */
+    auto run_key_passes = [&](const uint64_t **rk, const uint64_t **rv) -> hipError_t {
+        int plist[8], np = 0;
+        for (int p = 0; p < 8; ++p)
+            if (pass_on[p]) plist[np++] = p;
+        if (np == 0) { *rk = in_k; *rv = in_v; return hipSuccess; }
+        const uint64_t *cur = nullptr;
+        for (int i = 0; i < np; ++i) {
+            int p = plist[i];
+            bool in_pk = i > 0, out_pk = i < np - 1;
+            uint64_t *dbuf = ((i & 1) == 0) ? pA : pB;
+            uint64_t *dk = dbuf;
+            uint64_t *dv = out_pk ? nullptr : dbuf + n;
             RadixDigit df{8 * p};
-            HIP_TRY(scatter_pass_osw(s, *ck, *cv, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, true, df, "radix_scatter"));
-            *ck = dk;
-            *cv = dv;
+            HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur, i == 0 ? in_v : nullptr,
+                                     n, gbase_d + p * 256, desc, ticket,
+                                     dk, dv, true, in_pk, out_pk, df, "radix_scatter"));
+            cur = dk;
         }
+        *rk = cur;
+        *rv = cur + n;
         return hipSuccess;
     };
 
@@ -868,9 +936,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         int active = 8;
         HIP_TRY(exact_hists(false, &active));
         if (active <= 5) { /* narrow keys: skipped key sort groups exactly */
-            HIP_TRY(run_key_passes(&cur_k, &cur_v));
-            *res_k = cur_k;
-            *res_v = cur_v;
+            HIP_TRY(run_key_passes(res_k, res_v));
             return hipSuccess;
         }
         /* sample lied; fall through to the hash path */
@@ -883,14 +949,17 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     const uint64_t hmask = (hbytes >= 8) ? ~0ULL : ((1ULL << (8 * hbytes)) - 1);
     int active5 = 0;
     HIP_TRY(exact_hists(true, &active5));
-    for (int p = 0; p < hbytes; ++p) {
-        uint64_t *dk = (cur_k == ak) ? bk : ak;
-        uint64_t *dv = (cur_v == av) ? bv : av;
-        HashByteDigit df{8 * p};
-        HIP_TRY(scatter_pass_osw(s, cur_k, cur_v, n, gbase_d + p * 256, desc, ticket,
-                                 dk, dv, true, df, "radix_scatter"));
+    for (int i = 0; i < hbytes; ++i) {
+        bool in_pk = i > 0, out_pk = i < hbytes - 1;
+        uint64_t *dbuf = ((i & 1) == 0) ? pA : pB;
+        uint64_t *dk = dbuf;
+        uint64_t *dv = out_pk ? nullptr : dbuf + n;
+        HashByteDigit df{8 * i};
+        HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur_k, i == 0 ? in_v : nullptr,
+                                 n, gbase_d + i * 256, desc, ticket,
+                                 dk, dv, true, in_pk, out_pk, df, "radix_scatter"));
         cur_k = dk;
-        cur_v = dv;
+        cur_v = out_pk ? nullptr : dk + n;
     }
     HIP_TRY(hipMemsetAsync(d_err, 0, 4, s));
     {
@@ -906,8 +975,6 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     if (err) { /* an oversized hash-dirty run: full key sort instead */
         int active = 8;
         HIP_TRY(exact_hists(false, &active));
-        cur_k = in_k;
-        cur_v = in_v;
         HIP_TRY(run_key_passes(&cur_k, &cur_v));
     }
     *res_k = cur_k;
