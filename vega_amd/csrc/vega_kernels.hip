@@ -472,44 +472,43 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
 typedef __attribute__((address_space(1))) unsigned long long gdesc_t;
 
 template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
-__global__ __launch_bounds__(BLOCK) void k_scatter_osw(
+__global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
     uint64_t *out_k, uint64_t *out_v, DF df) {
-    /* IN_PK/OUT_PK: rows are interleaved (k,v) ulonglong2 — one 16-B vector
-     * access per row instead of two 8-B streams. The first pass of a sort
-     * reads SoA and packs; the last unpacks back to SoA for free. */
+    /* 512 threads = 8 waves per block (16 waves/CU at 2 blocks): each wave
+     * ranks a 512-row chunk of the 4096-row tile. IN_PK/OUT_PK: interleaved
+     * (k,v) rows — one 16-B vector access per row. */
+    constexpr int SB = 512, SW = 8, SIPT = 8;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     ulonglong2 *spk = (ulonglong2 *)smem;                       /* TILE 16B */
-    uint64_t *sk = (uint64_t *)smem;                            /* no-vals: TILE u64 */
+    uint64_t *sk = (uint64_t *)smem;                            /* no-vals */
     uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 16 : 8) * (size_t)TILE);
-    uint32_t *whist = hist + 256;
-    uint32_t *wsc = whist + 4 * 256;
-    uint32_t *vbp = wsc + 8;
-    uint32_t *tilebase = vbp + 8;
+    uint32_t *whist = hist + 256;                               /* SW*256 */
+    uint32_t *wsc = whist + SW * 256;                           /* 8 */
+    uint32_t *vbp = wsc + 8;                                    /* 8 */
+    uint32_t *tilebase = vbp + 8;                               /* 256 */
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t lower = ((uint64_t)1 << lane) - 1;
 
     if (t == 0) vbp[0] = atomicAdd(ticket, 1u); /* scheduling-ordered tile id */
-    for (int i = t; i < 4 * 256; i += BLOCK) whist[i] = 0;
+    for (int i = t; i < SW * 256; i += SB) whist[i] = 0;
     __syncthreads();
     const uint32_t vb = vbp[0];
     const uint64_t tbase = (uint64_t)vb * TILE;
     const uint32_t tile_n = (uint32_t)((n - tbase < TILE) ? (n - tbase) : TILE);
 
-    /* prefetch the wave's whole 1024-row chunk first (independent loads in
-     * flight together: ONE memory latency for the chunk, not one per round
-     * — the ranking rounds below are then register-only + LDS). Interior
-     * chunks take the unguarded path so nothing blocks load pipelining. */
-    uint64_t kk[IPT], vv[IPT];
-    uint32_t rank[IPT];
-    uint16_t dd[IPT];
-    const uint64_t chunk_g = tbase + (uint64_t)w * (64 * IPT) + lane;
-    const bool chunk_full = tbase + ((uint64_t)w + 1) * (64 * IPT) <= n; /* wave-uniform */
+    /* prefetch the wave's whole 512-row chunk (independent loads in flight
+     * together: ONE memory latency per chunk, not one per round) */
+    uint64_t kk[SIPT], vv[SIPT];
+    uint32_t rank[SIPT];
+    uint16_t dd[SIPT];
+    const uint64_t chunk_g = tbase + (uint64_t)w * (64 * SIPT) + lane;
+    const bool chunk_full = tbase + ((uint64_t)w + 1) * (64 * SIPT) <= n;
     if (chunk_full) {
 #pragma unroll
-        for (int r = 0; r < IPT; ++r) {
+        for (int r = 0; r < SIPT; ++r) {
             uint64_t idx = chunk_g + (uint64_t)r * 64;
             if (IN_PK) {
                 ulonglong2 kv = ((const ulonglong2 *)in_k)[idx];
@@ -522,7 +521,7 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
         }
     } else {
 #pragma unroll
-        for (int r = 0; r < IPT; ++r) {
+        for (int r = 0; r < SIPT; ++r) {
             uint64_t idx = chunk_g + (uint64_t)r * 64;
             bool valid = idx < n;
             kk[r] = 0;
@@ -539,9 +538,9 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
             }
         }
     }
-    /* ranking: wave w ranks its chunk in 16 register-only rounds */
+    /* ranking: register-only rounds over the prefetched chunk */
 #pragma unroll
-    for (int r = 0; r < IPT; ++r) {
+    for (int r = 0; r < SIPT; ++r) {
         bool valid = chunk_g + (uint64_t)r * 64 < n;
         uint32_t d = valid ? df(kk[r]) : 0;
         uint64_t m = wave_match8(d, valid);
@@ -556,12 +555,20 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
     }
     __syncthreads();
 
-    /* block digit counts -> publish AGGREGATE early, then local starts */
-    uint32_t c0 = whist[t], c1 = whist[256 + t], c2 = whist[512 + t], c3 = whist[768 + t];
-    uint32_t cnt = c0 + c1 + c2 + c3;
-    __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
-                       (unsigned long long)cnt | OSW_ST_AGG,
-                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    /* digit threads (t<256): counts -> publish AGG, local starts, per-wave
+     * offsets, lookback */
+    uint32_t cw[SW];
+    uint32_t cnt = 0;
+    if (t < 256) {
+#pragma unroll
+        for (int wv = 0; wv < SW; ++wv) {
+            cw[wv] = whist[wv * 256 + t];
+            cnt += cw[wv];
+        }
+        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
+                           (unsigned long long)cnt | OSW_ST_AGG,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    }
     uint32_t inc = cnt;
     for (int off = 1; off < 64; off <<= 1) {
         uint32_t u = __shfl_up(inc, off);
@@ -569,44 +576,46 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
     }
     if (lane == 63) wsc[w] = inc;
     __syncthreads();
-    uint32_t excl = inc - cnt;
-    for (int i = 0; i < w; ++i) excl += wsc[i];
-    hist[t] = excl;
-    whist[t] = 0;
-    whist[256 + t] = c0;
-    whist[512 + t] = c0 + c1;
-    whist[768 + t] = c0 + c1 + c2;
-
-    /* decoupled lookback: resolve this digit's prefix over earlier tiles */
-    unsigned long long excl_tiles = 0;
-    if (vb > 0) {
-        uint64_t j = vb - 1;
-        for (;;) {
-            unsigned long long wv = __hip_atomic_load(
-                (gdesc_t *)&desc[j * 256 + t], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-            unsigned long long st = wv >> 62;
-            if (st == 2) { excl_tiles += wv & OSW_CNT_MASK; break; }
-            if (st == 1) {
-                excl_tiles += wv & OSW_CNT_MASK;
-                if (j == 0) break;
-                j--;
-            } else {
-                __builtin_amdgcn_s_sleep(1);
+    if (t < 256) {
+        uint32_t excl = inc - cnt;
+        for (int i = 0; i < w; ++i) excl += wsc[i];
+        hist[t] = excl;
+        uint32_t run = 0;
+#pragma unroll
+        for (int wv = 0; wv < SW; ++wv) {
+            whist[wv * 256 + t] = run;
+            run += cw[wv];
+        }
+        unsigned long long excl_tiles = 0;
+        if (vb > 0) {
+            uint64_t j = vb - 1;
+            for (;;) {
+                unsigned long long wv2 = __hip_atomic_load(
+                    (gdesc_t *)&desc[j * 256 + t], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                unsigned long long st = wv2 >> 62;
+                if (st == 2) { excl_tiles += wv2 & OSW_CNT_MASK; break; }
+                if (st == 1) {
+                    excl_tiles += wv2 & OSW_CNT_MASK;
+                    if (j == 0) break;
+                    j--;
+                } else {
+                    __builtin_amdgcn_s_sleep(1);
+                }
             }
         }
+        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
+                           (excl_tiles + cnt) | OSW_ST_INC,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
     }
-    __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
-                       (excl_tiles + cnt) | OSW_ST_INC,
-                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
     __syncthreads();
 
     /* reorder into LDS at the stable tile-local position */
     {
-        uint32_t chunk0 = (uint32_t)w * (64 * IPT);
+        uint32_t chunk0 = (uint32_t)w * (64 * SIPT);
         uint32_t chunk_n = tile_n > chunk0 ? tile_n - chunk0 : 0;
 #pragma unroll
-        for (int r = 0; r < IPT; ++r) {
+        for (int r = 0; r < SIPT; ++r) {
             uint32_t local = (uint32_t)r * 64 + lane;
             if (local < chunk_n) {
                 uint32_t d = dd[r];
@@ -619,7 +628,7 @@ __global__ __launch_bounds__(BLOCK) void k_scatter_osw(
     __syncthreads();
 
     /* write out: digit-contiguous global writes */
-    for (uint32_t p = t; p < tile_n; p += BLOCK) {
+    for (uint32_t p = t; p < tile_n; p += SB) {
         uint64_t k, v = 0;
         if (HAS_VALS) {
             ulonglong2 kv = spk[p];
@@ -650,21 +659,21 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
-    size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 4 * 256 + 8 + 8 + 256) * 4;
+    size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(BLOCK), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
                            in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, df);
     } else if (!in_pk && !out_pk) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(BLOCK), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
                            in_k, in_v, n, gbase_d, desc, ticket, out_k, out_v, df);
     } else if (!in_pk && out_pk) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(BLOCK), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
                            in_k, in_v, n, gbase_d, desc, ticket, out_k, nullptr, df);
     } else if (in_pk && out_pk) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(BLOCK), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
                            in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, df);
     } else {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(BLOCK), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
                            in_k, nullptr, n, gbase_d, desc, ticket, out_k, out_v, df);
     }
     return hipGetLastError();
@@ -817,32 +826,59 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
  * stay within a run, so every observed key keeps its run's h40. */
 __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, uint64_t n,
                                 uint64_t hmask, int *err) {
+    /* per-thread 8-row register window: each key hashed once; runs owned by
+     * the chunk holding their START (race-free: other chunks only compare
+     * hashes, which permutation within a run preserves) */
+    constexpr int C = 8;
+    uint64_t nchunks = (n + C - 1) / C;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        uint64_t ki = k[i];
-        uint64_t hi_ = vega_hash_u64(ki) & hmask;
-        if (i > 0 && (vega_hash_u64(k[i - 1]) & hmask) == hi_)
-            continue; /* not a run start */
-        uint64_t j = i + 1;
-        bool dirty = false;
-        while (j < n) {
-            uint64_t kj = k[j];
-            if ((vega_hash_u64(kj) & hmask) != hi_) break;
-            dirty |= (kj != ki);
-            j++;
+    for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
+        uint64_t base = c * C;
+        int m = (int)((n - base < C) ? (n - base) : C);
+        uint64_t kk[C], hh[C];
+        if (m == C) {
+#pragma unroll
+            for (int j = 0; j < C; ++j) kk[j] = k[base + j];
+        } else {
+            for (int j = 0; j < m; ++j) kk[j] = k[base + j];
         }
-        if (!dirty) continue;
-        if (j - i > 64) { *err = 1; continue; }
-        for (uint64_t x = i + 1; x < j; x++) {
-            uint64_t kx = k[x], vx = v[x];
-            uint64_t y = x;
-            while (y > i && k[y - 1] > kx) {
-                k[y] = k[y - 1];
-                v[y] = v[y - 1];
-                y--;
+#pragma unroll
+        for (int j = 0; j < C; ++j)
+            if (j < m) hh[j] = vega_hash_u64(kk[j]) & hmask;
+        uint64_t hprev = ~0ULL;
+        if (base > 0) hprev = vega_hash_u64(k[base - 1]) & hmask;
+        for (int j = 0; j < m;) {
+            bool is_start = (base + j == 0) || (hh[j] != (j > 0 ? hh[j - 1] : hprev));
+            if (!is_start) { j++; continue; }
+            uint64_t h0 = hh[j];
+            bool dirty = false;
+            int e = j + 1;
+            while (e < m && hh[e] == h0) { dirty |= (kk[e] != kk[j]); e++; }
+            uint64_t ge = base + e;
+            if (e == m) { /* run may continue beyond the window */
+                while (ge < n) {
+                    uint64_t kx = k[ge];
+                    if ((vega_hash_u64(kx) & hmask) != h0) break;
+                    dirty |= (kx != kk[j]);
+                    ge++;
+                }
             }
-            k[y] = kx;
-            v[y] = vx;
+            if (dirty) {
+                uint64_t gs = base + j;
+                if (ge - gs > 64) { *err = 1; j = e; continue; }
+                for (uint64_t x = gs + 1; x < ge; x++) {
+                    uint64_t kx = k[x], vx = v[x];
+                    uint64_t y = x;
+                    while (y > gs && k[y - 1] > kx) {
+                        k[y] = k[y - 1];
+                        v[y] = v[y - 1];
+                        y--;
+                    }
+                    k[y] = kx;
+                    v[y] = vx;
+                }
+            }
+            j = e;
         }
     }
 }
