@@ -826,59 +826,32 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
  * stay within a run, so every observed key keeps its run's h40. */
 __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, uint64_t n,
                                 uint64_t hmask, int *err) {
-    /* per-thread 8-row register window: each key hashed once; runs owned by
-     * the chunk holding their START (race-free: other chunks only compare
-     * hashes, which permutation within a run preserves) */
-    constexpr int C = 8;
-    uint64_t nchunks = (n + C - 1) / C;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
-        uint64_t base = c * C;
-        int m = (int)((n - base < C) ? (n - base) : C);
-        uint64_t kk[C], hh[C];
-        if (m == C) {
-#pragma unroll
-            for (int j = 0; j < C; ++j) kk[j] = k[base + j];
-        } else {
-            for (int j = 0; j < m; ++j) kk[j] = k[base + j];
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t ki = k[i];
+        uint64_t hi_ = vega_hash_u64(ki) & hmask;
+        if (i > 0 && (vega_hash_u64(k[i - 1]) & hmask) == hi_)
+            continue; /* not a run start */
+        uint64_t j = i + 1;
+        bool dirty = false;
+        while (j < n) {
+            uint64_t kj = k[j];
+            if ((vega_hash_u64(kj) & hmask) != hi_) break;
+            dirty |= (kj != ki);
+            j++;
         }
-#pragma unroll
-        for (int j = 0; j < C; ++j)
-            if (j < m) hh[j] = vega_hash_u64(kk[j]) & hmask;
-        uint64_t hprev = ~0ULL;
-        if (base > 0) hprev = vega_hash_u64(k[base - 1]) & hmask;
-        for (int j = 0; j < m;) {
-            bool is_start = (base + j == 0) || (hh[j] != (j > 0 ? hh[j - 1] : hprev));
-            if (!is_start) { j++; continue; }
-            uint64_t h0 = hh[j];
-            bool dirty = false;
-            int e = j + 1;
-            while (e < m && hh[e] == h0) { dirty |= (kk[e] != kk[j]); e++; }
-            uint64_t ge = base + e;
-            if (e == m) { /* run may continue beyond the window */
-                while (ge < n) {
-                    uint64_t kx = k[ge];
-                    if ((vega_hash_u64(kx) & hmask) != h0) break;
-                    dirty |= (kx != kk[j]);
-                    ge++;
-                }
+        if (!dirty) continue;
+        if (j - i > 64) { *err = 1; continue; }
+        for (uint64_t x = i + 1; x < j; x++) {
+            uint64_t kx = k[x], vx = v[x];
+            uint64_t y = x;
+            while (y > i && k[y - 1] > kx) {
+                k[y] = k[y - 1];
+                v[y] = v[y - 1];
+                y--;
             }
-            if (dirty) {
-                uint64_t gs = base + j;
-                if (ge - gs > 64) { *err = 1; j = e; continue; }
-                for (uint64_t x = gs + 1; x < ge; x++) {
-                    uint64_t kx = k[x], vx = v[x];
-                    uint64_t y = x;
-                    while (y > gs && k[y - 1] > kx) {
-                        k[y] = k[y - 1];
-                        v[y] = v[y - 1];
-                        y--;
-                    }
-                    k[y] = kx;
-                    v[y] = vx;
-                }
-            }
-            j = e;
+            k[y] = kx;
+            v[y] = vx;
         }
     }
 }
@@ -1105,15 +1078,41 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     uint32_t excl = inc - cnt;
     for (int i = 0; i < w; ++i) excl += wsc[i];
 
-    /* Walk the chunk. A run that both STARTS (its head) and ENDS (next head)
-     * inside this chunk is owned exclusively by this thread -> plain store
-     * (out_vv was zero/identity-initialised only for the boundary atomics).
-     * Runs crossing a chunk edge are flushed with device atomics (i64 adds
-     * wrap, so the result stays bit-exact regardless of flush order). */
+    /* Fast path: a fully-distinct interior chunk whose last run does not
+     * continue into the next chunk (the dominant C1 shape) emits its 16
+     * length-1 runs with pure vector stores — no branches, no atomics. */
+    bool fast = false;
+    if (cnt == IPT && c0g + IPT <= n) {
+        uint64_t nk = (c0g + IPT < n) ? k[c0g + IPT] : ~kk[IPT - 1];
+        fast = (nk != kk[IPT - 1]);
+    }
     int64_t segid = (int64_t)head_base[blockIdx.x] + excl - 1;
     int64_t acc_i = (OP == 3) ? INT64_MAX : (OP == 4) ? INT64_MIN : 0;
     double acc_f = 0.0;
     bool have = false, started_here = false;
+    if (fast) {
+        int64_t s0 = segid + 1;
+        int64_t *okp = (int64_t *)out_k + s0;
+        uint64_t *ovp = (uint64_t *)out_vv + s0;
+        if ((s0 & 1) == 0) {
+            ulonglong2 *ok2 = (ulonglong2 *)okp;
+            ulonglong2 *ov2 = (ulonglong2 *)ovp;
+#pragma unroll
+            for (int j = 0; j < IPT / 2; ++j) {
+                ok2[j] = make_ulonglong2(kk[2 * j], kk[2 * j + 1]);
+                if (OP == 1) ov2[j] = make_ulonglong2(1, 1);
+                else ov2[j] = make_ulonglong2(sv[2 * j], sv[2 * j + 1]);
+            }
+        } else {
+#pragma unroll
+            for (int j = 0; j < IPT; ++j) {
+                okp[j] = (int64_t)kk[j];
+                ovp[j] = (OP == 1) ? 1ULL : sv[j];
+            }
+        }
+        /* have stays false: the wave tail-combine below is inert for this
+         * lane but still participates in the shfl lanes */
+    } else
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
         uint64_t gi = c0g + j;
