@@ -184,15 +184,6 @@ __global__ void k_gen_uniform(int64_t *keys, int64_t *vals, uint64_t n,
     }
 }
 
-__global__ void k_xor_key_copy(const uint64_t *in_k, const uint64_t *in_v,
-                               uint64_t *out_k, uint64_t *out_v, uint64_t n,
-                               uint64_t xmask, int has_vals) {
-    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        out_k[i] = in_k[i] ^ xmask;
-        if (has_vals) out_v[i] = in_v[i];
-    }
-}
 
 __global__ void k_fill_i64(int64_t *p, uint64_t n, int64_t v) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
