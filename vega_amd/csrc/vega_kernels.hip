@@ -817,32 +817,64 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
  * stay within a run, so every observed key keeps its run's h40. */
 __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, uint64_t n,
                                 uint64_t hmask, int *err) {
+    /* two rows per thread (16-B aligned vector load), one hash per row; the
+     * global walk only happens when the next row's hash matches (rare at
+     * <=0.25 bucket load). Runs are owned by their START row's thread. */
+    uint64_t nchunks = (n + 1) / 2;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        uint64_t ki = k[i];
-        uint64_t hi_ = vega_hash_u64(ki) & hmask;
-        if (i > 0 && (vega_hash_u64(k[i - 1]) & hmask) == hi_)
-            continue; /* not a run start */
-        uint64_t j = i + 1;
-        bool dirty = false;
-        while (j < n) {
-            uint64_t kj = k[j];
-            if ((vega_hash_u64(kj) & hmask) != hi_) break;
-            dirty |= (kj != ki);
-            j++;
+    for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
+        uint64_t i0 = 2 * c;
+        uint64_t k0, k1 = 0;
+        bool has1 = i0 + 1 < n;
+        if (has1) {
+            ulonglong2 kv = ((const ulonglong2 *)k)[c];
+            k0 = kv.x;
+            k1 = kv.y;
+        } else {
+            k0 = k[i0];
         }
-        if (!dirty) continue;
-        if (j - i > 64) { *err = 1; continue; }
-        for (uint64_t x = i + 1; x < j; x++) {
-            uint64_t kx = k[x], vx = v[x];
-            uint64_t y = x;
-            while (y > i && k[y - 1] > kx) {
-                k[y] = k[y - 1];
-                v[y] = v[y - 1];
-                y--;
+        uint64_t h0 = vega_hash_u64(k0) & hmask;
+        uint64_t h1 = has1 ? (vega_hash_u64(k1) & hmask) : ~0ULL;
+        uint64_t hp = (i0 > 0) ? (vega_hash_u64(k[i0 - 1]) & hmask) : ~0ULL;
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+            uint64_t gi = i0 + j;
+            if (j == 1 && !has1) break;
+            uint64_t hj = j ? h1 : h0;
+            uint64_t kj = j ? k1 : k0;
+            uint64_t hprev = j ? h0 : hp;
+            bool is_start = (gi == 0) || (hj != hprev);
+            if (!is_start) continue;
+            /* in-window peek: run of length 1 needs no global walk */
+            uint64_t hnext = j ? ~0ULL : h1;
+            bool may_extend = (j == 1) || (hnext == hj) || (!has1 && gi + 1 < n);
+            if (j == 0 && has1 && hnext != hj) continue; /* len-1, clean */
+            (void)may_extend;
+            uint64_t je = gi + 1;
+            bool dirty = false;
+            if (j == 0 && has1) { /* h1 == h0 here */
+                dirty |= (k1 != kj);
+                je = gi + 2;
             }
-            k[y] = kx;
-            v[y] = vx;
+            while (je < n) {
+                uint64_t kx = k[je];
+                if ((vega_hash_u64(kx) & hmask) != hj) break;
+                dirty |= (kx != kj);
+                je++;
+            }
+            if (!dirty) continue;
+            if (je - gi > 64) { *err = 1; continue; }
+            for (uint64_t x = gi + 1; x < je; x++) {
+                uint64_t kx = k[x], vx = v[x];
+                uint64_t y = x;
+                while (y > gi && k[y - 1] > kx) {
+                    k[y] = k[y - 1];
+                    v[y] = v[y - 1];
+                    y--;
+                }
+                k[y] = kx;
+                v[y] = vx;
+            }
         }
     }
 }
