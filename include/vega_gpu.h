@@ -95,6 +95,10 @@ int vega_gpu_join(vega_ctx_t *ctx, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
 /* distinct (rdd.rs:501-531): keys of rdd deduplicated */
 int vega_gpu_distinct(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
                       vega_rdd_t *out);
+/* count_by_value (rdd.rs:449-459 = map(x->(x,1)) + reduce_by_key(+)):
+ * counts over the VALUE column; result rows are (value, count) */
+int vega_gpu_count_by_value(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
+                            vega_rdd_t *out);
 
 /* ---------------- actions ---------------- */
 int vega_gpu_count(vega_ctx_t *ctx, vega_rdd_t rdd, uint64_t *n);

@@ -150,6 +150,18 @@ def test_group_count_zipf(ctx):
     rdd.free(); red.free()
 
 
+def test_count_by_value_golden(ctx):
+    # test_pair_rdd.rs:85-109 via the dedicated API entry
+    vals = np.array([1, 2, 1, 3, 2, 3, 3, 2, 3], dtype=np.int64)
+    keys = np.zeros(len(vals), dtype=np.int64)
+    for parts in (4, 2):
+        rdd = ctx.make_rdd(keys, vals, nparts=parts)
+        cbv = rdd.count_by_value(nparts=parts)
+        gk, gv = cbv.collect()
+        assert sorted_pairs(gk, gv) == [(1, 2), (2, 3), (3, 4)]
+        rdd.free(); cbv.free()
+
+
 # ---------------- sort_by_key ----------------
 
 @pytest.mark.parametrize("n,bits,seed", [

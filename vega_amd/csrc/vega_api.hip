@@ -198,6 +198,30 @@ int vega_gpu_distinct(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts, vega_rdd_t
     return reduce_common(c, rdd, VEGA_OP_COUNT, nparts, out);
 }
 
+/* count_by_value (rdd.rs:449-459): group-count with the VALUE column as the
+ * key — the composition map(x->(x,1)) + reduce_by_key(+) collapses to the
+ * same sort+segmented-count the reduce path runs */
+int vega_gpu_count_by_value(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts,
+                            vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r || r->vtype != 0) return VEGA_ERR_INVALID;
+    int rc = ensure_ws(c, r->n);
+    if (rc) return rc;
+    RddImpl *o;
+    rc = new_rdd(c, r->n ? r->n : 1, 0, nparts, &o, out);
+    if (rc) return rc;
+    Ws ws(c->ws, c->ws_bytes);
+    const uint64_t *sk, *sv;
+    CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_v, (const uint64_t *)r->d_k,
+                              r->n, ws, &sk, &sv));
+    uint64_t nout = 0;
+    CTX_TRY(c, seg_reduce(c->stream, sk, sv, r->n, VEGA_OP_COUNT,
+                          (uint64_t *)o->d_k, o->d_v, &nout, ws));
+    o->n = nout;
+    return VEGA_OK;
+}
+
 int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
     RddImpl *r = get_rdd(c, rdd);

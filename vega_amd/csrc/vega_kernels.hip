@@ -466,7 +466,7 @@ template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
 __global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
-    uint64_t *out_k, uint64_t *out_v, DF df) {
+    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, DF df) {
     /* 512 threads = 8 waves per block (16 waves/CU at 2 blocks): each wave
      * ranks a 512-row chunk of the 4096-row tile. IN_PK/OUT_PK: interleaved
      * (k,v) rows — one 16-B vector access per row. */
@@ -635,6 +635,9 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         } else {
             out_k[gpos] = k;
             if (HAS_VALS) out_v[gpos] = v;
+            if (h32_out) /* low hash bits for the grouping cleanup (4 B/row
+                            beats its re-hashing 16 B/row) */
+                h32_out[gpos] = (uint32_t)vega_hash_u64(k);
         }
     }
 }
@@ -643,8 +646,8 @@ template <class DF>
 static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
                                    uint64_t n, const uint32_t *gbase_d,
                                    unsigned long long *desc, uint32_t *ticket,
-                                   uint64_t *out_k, uint64_t *out_v, bool has_vals,
-                                   bool in_pk, bool out_pk,
+                                   uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out,
+                                   bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
@@ -653,19 +656,19 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
         hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, df);
+                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
     } else if (!in_pk && !out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, gbase_d, desc, ticket, out_k, out_v, df);
+                           in_k, in_v, n, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
     } else if (!in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, gbase_d, desc, ticket, out_k, nullptr, df);
+                           in_k, in_v, n, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
     } else if (in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, df);
+                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
     } else {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, out_v, df);
+                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
     }
     return hipGetLastError();
 }
@@ -793,11 +796,11 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         if (p == 7 && signed_order) {
             RadixDigitTopSigned df{56};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, has_vals, in_pk, out_pk, df, "radix_scatter"));
         } else {
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, has_vals, in_pk, out_pk, df, "radix_scatter"));
         }
         cur = dk;
     }
@@ -815,53 +818,40 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
  * length 1 + n/2^40); a dirty run longer than 64 sets *err and the caller
  * falls back to the full key sort. Cross-run reads are safe: permutations
  * stay within a run, so every observed key keeps its run's h40. */
-__global__ void k_group_cleanup(uint64_t *k, uint64_t *v, uint64_t n,
-                                uint64_t hmask, int *err) {
-    /* two rows per thread (16-B aligned vector load), one hash per row; the
-     * global walk only happens when the next row's hash matches (rare at
-     * <=0.25 bucket load). Runs are owned by their START row's thread. */
-    uint64_t nchunks = (n + 1) / 2;
+/* cleanup v4: run boundaries come from the h32 side array the last hash
+ * pass wrote (4 B/row sequential reads instead of re-hashing 16 B/row).
+ * Runs are EQUAL-h32 groups — a superset of the hmask runs, so sorting a
+ * combined group by key still leaves equal keys adjacent (all the grouping
+ * contract needs). h32 is never rewritten, so cross-thread reads stay
+ * consistent while a run's owner permutes k/v (keys within a run keep the
+ * run's h32 by definition). */
+__global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
+                                uint64_t n, int *err) {
+    uint64_t nchunks = (n + 3) / 4;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
-        uint64_t i0 = 2 * c;
-        uint64_t k0, k1 = 0;
-        bool has1 = i0 + 1 < n;
-        if (has1) {
-            ulonglong2 kv = ((const ulonglong2 *)k)[c];
-            k0 = kv.x;
-            k1 = kv.y;
+        uint64_t i0 = 4 * c;
+        int m = (int)((n - i0 < 4) ? (n - i0) : 4);
+        uint32_t hh[5];
+        if (m == 4) {
+            uint4 hv = ((const uint4 *)h32)[c];
+            hh[1] = hv.x; hh[2] = hv.y; hh[3] = hv.z; hh[4] = hv.w;
         } else {
-            k0 = k[i0];
+            for (int j = 0; j < m; ++j) hh[j + 1] = h32[i0 + j];
         }
-        uint64_t h0 = vega_hash_u64(k0) & hmask;
-        uint64_t h1 = has1 ? (vega_hash_u64(k1) & hmask) : ~0ULL;
-        uint64_t hp = (i0 > 0) ? (vega_hash_u64(k[i0 - 1]) & hmask) : ~0ULL;
-#pragma unroll
-        for (int j = 0; j < 2; ++j) {
+        hh[0] = (i0 > 0) ? h32[i0 - 1] : ~hh[1]; /* sentinel differs */
+        for (int j = 0; j < m; ++j) {
             uint64_t gi = i0 + j;
-            if (j == 1 && !has1) break;
-            uint64_t hj = j ? h1 : h0;
-            uint64_t kj = j ? k1 : k0;
-            uint64_t hprev = j ? h0 : hp;
-            bool is_start = (gi == 0) || (hj != hprev);
-            if (!is_start) continue;
-            /* in-window peek: run of length 1 needs no global walk */
-            uint64_t hnext = j ? ~0ULL : h1;
-            bool may_extend = (j == 1) || (hnext == hj) || (!has1 && gi + 1 < n);
-            if (j == 0 && has1 && hnext != hj) continue; /* len-1, clean */
-            (void)may_extend;
+            if (gi != 0 && hh[j + 1] == hh[j]) continue; /* not a run start */
+            /* peek in-window: single-row runs need no global traffic */
+            if (j + 1 < m && hh[j + 2] != hh[j + 1]) continue;
             uint64_t je = gi + 1;
+            while (je < n && h32[je] == hh[j + 1]) je++;
+            if (je - gi == 1) continue;
+            /* run of length >= 2: check keys */
+            uint64_t k0 = k[gi];
             bool dirty = false;
-            if (j == 0 && has1) { /* h1 == h0 here */
-                dirty |= (k1 != kj);
-                je = gi + 2;
-            }
-            while (je < n) {
-                uint64_t kx = k[je];
-                if ((vega_hash_u64(kx) & hmask) != hj) break;
-                dirty |= (kx != kj);
-                je++;
-            }
+            for (uint64_t x = gi + 1; x < je && !dirty; x++) dirty = (k[x] != k0);
             if (!dirty) continue;
             if (je - gi > 64) { *err = 1; continue; }
             for (uint64_t x = gi + 1; x < je; x++) {
@@ -898,7 +888,8 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
-    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err)
+    uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !h32buf)
         return hipErrorOutOfMemory;
 
     static thread_local uint32_t hh[8 * 256];
@@ -951,7 +942,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur, i == 0 ? in_v : nullptr,
                                      n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, true, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, true, in_pk, out_pk, df, "radix_scatter"));
             cur = dk;
         }
         *rk = cur;
@@ -978,7 +969,6 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
      * factor <= ~0.25 (4 bytes up to 2^30 rows, 5 above) — fewer passes,
      * slightly busier cleanup */
     const int hbytes = (n <= (1ULL << 30)) ? 4 : 5;
-    const uint64_t hmask = (hbytes >= 8) ? ~0ULL : ((1ULL << (8 * hbytes)) - 1);
     int active5 = 0;
     HIP_TRY(exact_hists(true, &active5));
     for (int i = 0; i < hbytes; ++i) {
@@ -989,7 +979,8 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HashByteDigit df{8 * i};
         HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur_k, i == 0 ? in_v : nullptr,
                                  n, gbase_d + i * 256, desc, ticket,
-                                 dk, dv, true, in_pk, out_pk, df, "radix_scatter"));
+                                 dk, dv, out_pk ? nullptr : h32buf,
+                                 true, in_pk, out_pk, df, "radix_scatter"));
         cur_k = dk;
         cur_v = out_pk ? nullptr : dk + n;
     }
@@ -998,7 +989,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         ProfScope ps("group_cleanup", s);
         uint32_t gb = nb < 2048 ? nb : 2048;
         hipLaunchKernelGGL(k_group_cleanup, dim3(gb), dim3(BLOCK), 0, s,
-                           (uint64_t *)cur_k, (uint64_t *)cur_v, n, hmask, d_err);
+                           (uint64_t *)cur_k, (uint64_t *)cur_v, h32buf, n, d_err);
         HIP_TRY(hipGetLastError());
     }
     int err = 0;
@@ -1326,6 +1317,7 @@ size_t ws_bytes_for(uint64_t n) {
     b += 4 * ((n * 8 + 255) & ~255ULL);           /* sort ping-pong k+v */
     b += ((size_t)256 * nb * 4 + 255) & ~255ULL;  /* bh matrix */
     b += ((size_t)256 * nb * 4 + 255) & ~255ULL;  /* raw block hists (transient) */
+    b += (n * 4 + 255) & ~255ULL;                 /* h32 side array (grouping cleanup) */
     b += 8 * 256 * 4 + 256;                       /* hist8 */
     b += (((size_t)nb + 2) * 4 + 255) & ~255ULL;  /* head counts */
     /* scan recursion partials: nb/TILE + nb/TILE^2 + ... < nb/2048 */

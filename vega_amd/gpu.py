@@ -145,6 +145,13 @@ class Rdd:
         return Rdd(self.ctx, out.value,
                    np.float64 if op == OP_SUM_F64 else np.int64)
 
+    def count_by_value(self, nparts=256):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_count_by_value(self.ctx._c, ctypes.c_uint64(self.h),
+                                             ctypes.c_uint32(nparts), ctypes.byref(out)),
+               "count_by_value", self.ctx._c)
+        return Rdd(self.ctx, out.value, np.int64)
+
     def group_count(self, nparts=256):
         out = ctypes.c_uint64()
         _check(lib().vega_gpu_group_count(self.ctx._c, ctypes.c_uint64(self.h),
