@@ -43,16 +43,22 @@ int main(int argc, char **argv) {
     try {
         vega::Context sc;
 
-        /* count_by_value golden (test_pair_rdd.rs:85-109): map v->(v,1) +
-         * reduce_by_key(+, splits) per rdd.rs:449-459 */
+        /* count_by_value golden (test_pair_rdd.rs:85-109): both as the
+         * explicit map v->(v,1) + reduce_by_key composition (rdd.rs:449-459)
+         * and through the dedicated count_by_value entry */
         {
             pairs_t in;
             for (int64_t x : {1, 2, 1, 3, 2, 3, 3, 2, 3}) in.push_back({x, 1});
             for (uint32_t parts : {4u, 2u}) {
                 auto r = sc.make_rdd(in, parts).reduce_by_key(VEGA_OP_SUM_I64, parts);
                 CHECK_EQ(sorted(r.collect()), (pairs_t{{1, 2}, {2, 3}, {3, 4}}),
-                         "count_by_value golden");
+                         "count_by_value golden (composition)");
             }
+            pairs_t in2;
+            for (int64_t x : {1, 2, 1, 3, 2, 3, 3, 2, 3}) in2.push_back({0, x});
+            auto r2 = sc.make_rdd(in2, 4).count_by_value(4);
+            CHECK_EQ(sorted(r2.collect()), (pairs_t{{1, 2}, {2, 3}, {3, 4}}),
+                     "count_by_value golden (API)");
         }
         /* group_by_key golden counts (test_pair_rdd.rs:9-37; x->120, y->121) */
         {

@@ -65,6 +65,12 @@ class PairRdd {
         check(vega_gpu_join(c_, h_, other.h_, num_splits, &out), "join", c_);
         return PairRdd(c_, out, false);
     }
+    /* rdd.rs:449-459: counts over the VALUE column */
+    PairRdd count_by_value(uint32_t num_splits = 256) const {
+        vega_rdd_t out = 0;
+        check(vega_gpu_count_by_value(c_, h_, num_splits, &out), "count_by_value", c_);
+        return PairRdd(c_, out, false);
+    }
     /* rdd.rs:501-531 */
     PairRdd distinct(uint32_t num_splits = 256) const {
         vega_rdd_t out = 0;
