@@ -464,7 +464,7 @@ typedef __attribute__((address_space(1))) unsigned long long gdesc_t;
 
 template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
 __global__ __launch_bounds__(512) void k_scatter_osw(
-    const uint64_t *in_k, const uint64_t *in_v, uint64_t n,
+    const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
     uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, DF df) {
     /* 512 threads = 8 waves per block (16 waves/CU at 2 blocks): each wave
@@ -556,7 +556,9 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
             cw[wv] = whist[wv * 256 + t];
             cnt += cw[wv];
         }
-        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
+        /* digit-major descriptors: each digit's chain is contiguous, so the
+         * lookback walk is cache-local and probes 4 predecessors at once */
+        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)t * nblocks + vb],
                            (unsigned long long)cnt | OSW_ST_AGG,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
@@ -579,22 +581,30 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         }
         unsigned long long excl_tiles = 0;
         if (vb > 0) {
-            uint64_t j = vb - 1;
-            for (;;) {
-                unsigned long long wv2 = __hip_atomic_load(
-                    (gdesc_t *)&desc[j * 256 + t], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                unsigned long long st = wv2 >> 62;
-                if (st == 2) { excl_tiles += wv2 & OSW_CNT_MASK; break; }
-                if (st == 1) {
-                    excl_tiles += wv2 & OSW_CNT_MASK;
-                    if (j == 0) break;
-                    j--;
-                } else {
-                    __builtin_amdgcn_s_sleep(1);
+            gdesc_t *col = (gdesc_t *)(desc + (uint64_t)t * nblocks);
+            int64_t j = (int64_t)vb - 1;
+            while (j >= 0) {
+                /* probe up to 4 predecessors with independent loads */
+                unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
+                int navail = (j >= 3) ? 4 : (int)(j + 1);
+                d0 = __hip_atomic_load(col + j, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 1) d1 = __hip_atomic_load(col + j - 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 2) d2 = __hip_atomic_load(col + j - 2, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 3) d3 = __hip_atomic_load(col + j - 3, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                bool done = false, stall = false;
+                unsigned long long dd4[4] = {d0, d1, d2, d3};
+                for (int q = 0; q < navail; ++q) {
+                    unsigned long long st = dd4[q] >> 62;
+                    if (st == 2) { excl_tiles += dd4[q] & OSW_CNT_MASK; done = true; break; }
+                    if (st == 1) { excl_tiles += dd4[q] & OSW_CNT_MASK; j--; continue; }
+                    stall = true;
+                    break;
                 }
+                if (done) break;
+                if (stall) __builtin_amdgcn_s_sleep(1);
             }
         }
-        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
+        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)t * nblocks + vb],
                            (excl_tiles + cnt) | OSW_ST_INC,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
@@ -656,19 +666,19 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
         hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
     } else if (!in_pk && !out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
     } else if (!in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
     } else if (in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
     } else {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
     }
     return hipGetLastError();
 }
