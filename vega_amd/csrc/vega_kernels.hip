@@ -556,9 +556,10 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
             cw[wv] = whist[wv * 256 + t];
             cnt += cw[wv];
         }
-        /* digit-major descriptors: each digit's chain is contiguous, so the
-         * lookback walk is cache-local and probes 4 predecessors at once */
-        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)t * nblocks + vb],
+        /* tile-major descriptors: a tile's 256 publishes are one contiguous
+         * 2 KB burst (digit-major measured 3.6x slower — 256 scattered
+         * lines per publish) */
+        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            (unsigned long long)cnt | OSW_ST_AGG,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
@@ -581,16 +582,16 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         }
         unsigned long long excl_tiles = 0;
         if (vb > 0) {
-            gdesc_t *col = (gdesc_t *)(desc + (uint64_t)t * nblocks);
+            gdesc_t *col = (gdesc_t *)(desc + t);
             int64_t j = (int64_t)vb - 1;
             while (j >= 0) {
                 /* probe up to 4 predecessors with independent loads */
                 unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
                 int navail = (j >= 3) ? 4 : (int)(j + 1);
-                d0 = __hip_atomic_load(col + j, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 1) d1 = __hip_atomic_load(col + j - 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 2) d2 = __hip_atomic_load(col + j - 2, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 3) d3 = __hip_atomic_load(col + j - 3, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                d0 = __hip_atomic_load(col + j * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 1) d1 = __hip_atomic_load(col + (j - 1) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 2) d2 = __hip_atomic_load(col + (j - 2) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 3) d3 = __hip_atomic_load(col + (j - 3) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 bool done = false, stall = false;
                 unsigned long long dd4[4] = {d0, d1, d2, d3};
                 for (int q = 0; q < navail; ++q) {
@@ -604,7 +605,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 if (stall) __builtin_amdgcn_s_sleep(1);
             }
         }
-        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)t * nblocks + vb],
+        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            (excl_tiles + cnt) | OSW_ST_INC,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
