@@ -104,6 +104,8 @@ def main():
                          "semantics (with map-side pre-combine before the "
                          "exchange); sort = C3 sort_by_key")
     ap.add_argument("--dist", choices=["uniform", "zipf"], default="uniform")
+    ap.add_argument("--dtype", choices=["i64", "f64"], default="i64",
+                    help="value type for --op reduce (f64 = SURVEY C1's f64-sum variant)")
     ap.add_argument("--zipf-keyspace", type=int, default=100_000_000)
     args = ap.parse_args()
 
@@ -132,6 +134,11 @@ def main():
                                     keyspace=args.zipf_keyspace, start=rank * rows)
         k = torch.from_numpy(hk).to(dev)
         v = torch.from_numpy(hv).to(dev)
+    elif args.dtype == "f64":
+        k = torch.empty(rows, dtype=torch.int64, device=dev)
+        v = torch.empty(rows, dtype=torch.float64, device=dev)
+        gpu.dev_gen_uniform_f64(k, v, seed=args.seed, key_bits=args.key_bits,
+                                start=rank * rows)
     else:
         k = torch.empty(rows, dtype=torch.int64, device=dev)
         v = torch.empty(rows, dtype=torch.int64, device=dev)
@@ -221,7 +228,8 @@ def main():
             gpu.dev_sort_pairs(rbk, rbv, ws)
             nout = gpu.dev_join_sorted(rak, rav, rbk, rbv, jk, jva, jvb, ws)
 
-    step = {"reduce": lambda: step_reduce(gpu.OP_SUM_I64),
+    step = {"reduce": lambda: step_reduce(
+                gpu.OP_SUM_F64 if args.dtype == "f64" else gpu.OP_SUM_I64),
             "group_count": step_group_count,
             "sort": step_sort,
             "join": step_join}[args.op]
@@ -292,7 +300,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": "int64",
+            "dtype": "float64" if args.dtype == "f64" else "int64",
             "data": "synthetic",
             "config": {
                 "workload": WORKLOAD if args.op == "reduce" and args.dist == "uniform"

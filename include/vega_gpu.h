@@ -127,6 +127,11 @@ size_t vega_dev_ws_bytes(uint64_t n);
 int vega_dev_gen_uniform_i64(void *stream, int64_t *keys, int64_t *vals,
                              uint64_t n, uint64_t seed, int key_bits,
                              uint64_t start);
+/* f64-value variant (values exact dyadic uniform [0,1), bit-identical to
+ * datagen.c's vega_gen_uniform_pairs_f64) */
+int vega_dev_gen_uniform_f64(void *stream, int64_t *keys, double *vals,
+                             uint64_t n, uint64_t seed, int key_bits,
+                             uint64_t start);
 
 /* map-side radix partition (replaces the per-row get_partition + HashMap of
  * dependency.rs:191-210): bucket of row = splitmix64(key) % nparts; rows

@@ -78,6 +78,18 @@ def test_reduce_by_key_empty(ctx):
     rdd.free(); red.free()
 
 
+def test_dev_gen_f64_matches_host(ctx):
+    import torch
+    from vega_amd import gpu
+    n = 50_000
+    k = torch.empty(n, dtype=torch.int64, device="cuda")
+    v = torch.empty(n, dtype=torch.float64, device="cuda")
+    gpu.dev_gen_uniform_f64(k, v, seed=19, key_bits=20)
+    hk, hv = datagen.uniform_pairs_f64(19, n, key_bits=20)
+    assert (k.cpu().numpy() == hk).all()
+    assert (v.cpu().numpy() == hv).all()  # exact dyadic conversion: bit-equal
+
+
 def test_reduce_by_key_f64(ctx):
     from vega_amd import gpu
     n = 500_000

@@ -153,7 +153,7 @@ int vega_gpu_gen_rdd_uniform(vega_ctx_t *c, uint64_t n, uint64_t seed, int key_b
     int rc = new_rdd(c, n ? n : 1, 0, nparts, &r, out);
     if (rc) return rc;
     r->n = n;
-    CTX_TRY(c, gen_uniform(c->stream, r->d_k, (int64_t *)r->d_v, n, seed, key_bits, start));
+    CTX_TRY(c, gen_uniform(c->stream, r->d_k, (int64_t *)r->d_v, n, seed, key_bits, start, false));
     return VEGA_OK;
 }
 
@@ -364,7 +364,15 @@ int vega_prof_stats(char *buf, size_t len) { return prof_stats_json(buf, len) < 
 
 int vega_dev_gen_uniform_i64(void *stream, int64_t *keys, int64_t *vals, uint64_t n,
                              uint64_t seed, int key_bits, uint64_t start) {
-    return gen_uniform((hipStream_t)stream, keys, vals, n, seed, key_bits, start)
+    return gen_uniform((hipStream_t)stream, keys, vals, n, seed, key_bits, start, false)
+               == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
+}
+
+/* f64-value variant of the generator (values exact dyadic uniform [0,1),
+ * bit-identical to datagen.c's vega_gen_uniform_pairs_f64) */
+int vega_dev_gen_uniform_f64(void *stream, int64_t *keys, double *vals, uint64_t n,
+                             uint64_t seed, int key_bits, uint64_t start) {
+    return gen_uniform((hipStream_t)stream, keys, (int64_t *)vals, n, seed, key_bits, start, true)
                == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
 }
 

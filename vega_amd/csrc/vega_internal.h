@@ -77,7 +77,7 @@ hipError_t range_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *
                            uint64_t *out_k, uint64_t *out_v, uint64_t *h_counts, Ws &ws);
 
 hipError_t gen_uniform(hipStream_t s, int64_t *keys, int64_t *vals, uint64_t n,
-                       uint64_t seed, int key_bits, uint64_t start);
+                       uint64_t seed, int key_bits, uint64_t start, bool f64_vals);
 
 hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
                           uint64_t n, uint64_t *h_sum, Ws &ws);

@@ -175,12 +175,17 @@ __device__ __forceinline__ uint64_t wave_match8(uint32_t d, bool valid) {
 /* generator / elementwise                                             */
 
 __global__ void k_gen_uniform(int64_t *keys, int64_t *vals, uint64_t n,
-                              uint64_t seed, uint64_t mask, uint64_t start) {
+                              uint64_t seed, uint64_t mask, uint64_t start,
+                              int f64_vals) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         uint64_t j = start + i;
         keys[i] = (int64_t)(vega_rand_u64(seed, 2 * j) & mask);
-        vals[i] = (int64_t)vega_rand_u64(seed, 2 * j + 1);
+        uint64_t r = vega_rand_u64(seed, 2 * j + 1);
+        if (f64_vals) /* exact dyadic [0,1): matches datagen.c f64 variant */
+            ((double *)vals)[i] = (double)(r >> 11) * 0x1p-53;
+        else
+            vals[i] = (int64_t)r;
     }
 }
 
@@ -1300,12 +1305,13 @@ hipError_t range_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *
 /* misc host entries                                                   */
 
 hipError_t gen_uniform(hipStream_t s, int64_t *keys, int64_t *vals, uint64_t n,
-                       uint64_t seed, int key_bits, uint64_t start) {
+                       uint64_t seed, int key_bits, uint64_t start, bool f64_vals) {
     uint64_t mask = (key_bits >= 64) ? ~0ULL : ((1ULL << key_bits) - 1);
     uint32_t nb = nblocks_for(n);
     uint32_t gb = nb < 2048 ? (nb ? nb : 1) : 2048;
     ProfScope ps("gen", s);
-    hipLaunchKernelGGL(k_gen_uniform, dim3(gb), dim3(BLOCK), 0, s, keys, vals, n, seed, mask, start);
+    hipLaunchKernelGGL(k_gen_uniform, dim3(gb), dim3(BLOCK), 0, s, keys, vals, n, seed,
+                       mask, start, f64_vals ? 1 : 0);
     return hipGetLastError();
 }
 

@@ -232,6 +232,14 @@ def dev_gen_uniform(keys_t, vals_t, seed, key_bits=63, start=0):
            "dev_gen")
 
 
+def dev_gen_uniform_f64(keys_t, vals_t, seed, key_bits=63, start=0):
+    n = keys_t.numel()
+    _check(lib().vega_dev_gen_uniform_f64(_stream(), _t(keys_t), _t(vals_t),
+                                          ctypes.c_uint64(n), ctypes.c_uint64(seed),
+                                          ctypes.c_int(key_bits), ctypes.c_uint64(start)),
+           "dev_gen_f64")
+
+
 def dev_partition(keys_t, vals_t, nparts, out_k, out_v, ws_t):
     n = keys_t.numel()
     counts = np.zeros(nparts, dtype=np.uint64)
