@@ -52,6 +52,22 @@ typedef enum {
     VEGA_OP_MAX_I64 = 4,
 } vega_op_t;
 
+/* narrow transforms (rdd.rs:199-235 map/filter, pair_rdd.rs:84-101
+ * map_values) as fixed op-enums — the device analogue of the reference's
+ * closures (like vega_op_t for the Aggregator). Outputs stay device-resident
+ * and feed the shuffle with no host round-trip. */
+typedef enum {
+    VEGA_MAP_VALUES_ADD = 0, /* v' = v + p0 (wrapping) */
+    VEGA_MAP_VALUES_MUL = 1, /* v' = v * p0 (wrapping) */
+    VEGA_MAP_KEYS_ADD   = 2, /* k' = k + p0 (wrapping) */
+    VEGA_MAP_SWAP       = 3, /* (k,v) -> (v,k) */
+} vega_map_op_t;
+typedef enum {
+    VEGA_PRED_KEY_MOD_EQ   = 0, /* keep if k % p0 == p1 (C signed rem) */
+    VEGA_PRED_VAL_GT       = 1, /* keep if v > p0 */
+    VEGA_PRED_KEY_IN_RANGE = 2, /* keep if p0 <= k < p1 */
+} vega_pred_t;
+
 typedef struct vega_ctx vega_ctx_t;
 typedef uint64_t vega_rdd_t;      /* opaque RDD handle, 0 = invalid */
 
@@ -99,6 +115,12 @@ int vega_gpu_distinct(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
  * counts over the VALUE column; result rows are (value, count) */
 int vega_gpu_count_by_value(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
                             vega_rdd_t *out);
+
+/* narrow transforms (device-resident; stable order for filter) */
+int vega_gpu_map(vega_ctx_t *ctx, vega_rdd_t rdd, vega_map_op_t op, int64_t p0,
+                 vega_rdd_t *out);
+int vega_gpu_filter(vega_ctx_t *ctx, vega_rdd_t rdd, vega_pred_t pred,
+                    int64_t p0, int64_t p1, vega_rdd_t *out);
 
 /* ---------------- actions ---------------- */
 int vega_gpu_count(vega_ctx_t *ctx, vega_rdd_t rdd, uint64_t *n);

@@ -307,6 +307,38 @@ int vega_gpu_collect_join(vega_ctx_t *c, vega_rdd_t rdd, int64_t *keys,
     return VEGA_OK;
 }
 
+int vega_gpu_map(vega_ctx_t *c, vega_rdd_t rdd, vega_map_op_t op, int64_t p0,
+                 vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r || r->vtype != 0) return VEGA_ERR_INVALID;
+    RddImpl *o;
+    int rc = new_rdd(c, r->n ? r->n : 1, 0, r->nparts, &o, out);
+    if (rc) return rc;
+    o->n = r->n;
+    CTX_TRY(c, narrow_map(c->stream, r->d_k, (const int64_t *)r->d_v, r->n,
+                          (int)op, p0, o->d_k, (int64_t *)o->d_v));
+    return VEGA_OK;
+}
+
+int vega_gpu_filter(vega_ctx_t *c, vega_rdd_t rdd, vega_pred_t pred,
+                    int64_t p0, int64_t p1, vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r || r->vtype != 0) return VEGA_ERR_INVALID;
+    int rc = ensure_ws(c, r->n);
+    if (rc) return rc;
+    RddImpl *o;
+    rc = new_rdd(c, r->n ? r->n : 1, 0, r->nparts, &o, out);
+    if (rc) return rc;
+    Ws ws(c->ws, c->ws_bytes);
+    uint64_t nout = 0;
+    CTX_TRY(c, narrow_filter(c->stream, r->d_k, (const int64_t *)r->d_v, r->n,
+                             (int)pred, p0, p1, o->d_k, (int64_t *)o->d_v, &nout, ws));
+    o->n = nout;
+    return VEGA_OK;
+}
+
 int vega_gpu_count(vega_ctx_t *c, vega_rdd_t rdd, uint64_t *n) {
     RddImpl *r = get_rdd(c, rdd);
     if (!r) return VEGA_ERR_INVALID;

@@ -79,6 +79,12 @@ hipError_t range_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *
 hipError_t gen_uniform(hipStream_t s, int64_t *keys, int64_t *vals, uint64_t n,
                        uint64_t seed, int key_bits, uint64_t start, bool f64_vals);
 
+hipError_t narrow_map(hipStream_t s, const int64_t *in_k, const int64_t *in_v,
+                      uint64_t n, int op, int64_t p0, int64_t *out_k, int64_t *out_v);
+hipError_t narrow_filter(hipStream_t s, const int64_t *in_k, const int64_t *in_v,
+                         uint64_t n, int pred, int64_t p0, int64_t p1,
+                         int64_t *out_k, int64_t *out_v, uint64_t *h_nout, Ws &ws);
+
 hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
                           uint64_t n, uint64_t *h_sum, Ws &ws);
 

@@ -1302,6 +1302,142 @@ hipError_t range_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *
 }
 
 /* ------------------------------------------------------------------ */
+/* narrow device ops (rdd.rs:199-235 map/filter, pair_rdd.rs:84-101
+ * map_values) as fixed op-enums — outputs stay device-resident and feed the
+ * shuffle with no host round-trip */
+
+__global__ void k_map_pairs(const int64_t *in_k, const int64_t *in_v, uint64_t n,
+                            int op, int64_t p0, int64_t *out_k, int64_t *out_v) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        int64_t k = in_k[i], v = in_v[i];
+        switch (op) {
+        case 0: v = (int64_t)((uint64_t)v + (uint64_t)p0); break; /* MAP_VALUES_ADD */
+        case 1: v = (int64_t)((uint64_t)v * (uint64_t)p0); break; /* MAP_VALUES_MUL */
+        case 2: k = (int64_t)((uint64_t)k + (uint64_t)p0); break; /* MAP_KEYS_ADD */
+        case 3: { int64_t t = k; k = v; v = t; break; }           /* MAP_SWAP */
+        default: break;
+        }
+        out_k[i] = k;
+        out_v[i] = v;
+    }
+}
+
+__device__ __forceinline__ bool vega_pred_eval(int pred, int64_t k, int64_t v,
+                                               int64_t p0, int64_t p1) {
+    switch (pred) {
+    case 0: return p0 != 0 && (k % p0) == p1; /* KEY_MOD_EQ */
+    case 1: return v > p0;                    /* VAL_GT */
+    case 2: return k >= p0 && k < p1;         /* KEY_IN_RANGE */
+    default: return false;
+    }
+}
+
+__global__ void k_filter_count(const int64_t *keys, const int64_t *vals, uint64_t n,
+                               int pred, int64_t p0, int64_t p1, uint32_t *bc) {
+    __shared__ uint32_t wsum[BLOCK / 64];
+    uint64_t tbase = (uint64_t)blockIdx.x * TILE;
+    uint32_t c = 0;
+#pragma unroll
+    for (int j = 0; j < IPT; ++j) {
+        uint64_t idx = tbase + (uint64_t)j * BLOCK + threadIdx.x;
+        if (idx < n) c += vega_pred_eval(pred, keys[idx], vals[idx], p0, p1);
+    }
+    for (int off = 32; off > 0; off >>= 1) c += __shfl_down(c, off);
+    int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+    if (lane == 0) wsum[w] = c;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t t = 0;
+        for (int i = 0; i < BLOCK / 64; ++i) t += wsum[i];
+        bc[blockIdx.x] = t;
+    }
+}
+
+/* stable compaction: per-thread chunks keep row order (same block-scan
+ * pattern as k_seg_emit) */
+__global__ void k_filter_emit(const int64_t *keys, const int64_t *vals, uint64_t n,
+                              int pred, int64_t p0, int64_t p1,
+                              const uint32_t *base, int64_t *out_k, int64_t *out_v) {
+    __shared__ uint32_t wsc[BLOCK / 64];
+    const int t = threadIdx.x, lane = t & 63, w = t >> 6;
+    const uint64_t c0g = (uint64_t)blockIdx.x * TILE + (uint64_t)t * IPT;
+    bool keep[IPT];
+    int64_t kk[IPT], vv[IPT];
+    uint32_t cnt = 0;
+#pragma unroll
+    for (int j = 0; j < IPT; ++j) {
+        uint64_t gi = c0g + j;
+        keep[j] = false;
+        if (gi < n) {
+            kk[j] = keys[gi];
+            vv[j] = vals[gi];
+            keep[j] = vega_pred_eval(pred, kk[j], vv[j], p0, p1);
+            cnt += keep[j];
+        }
+    }
+    uint32_t inc = cnt;
+    for (int off = 1; off < 64; off <<= 1) {
+        uint32_t u = __shfl_up(inc, off);
+        if (lane >= off) inc += u;
+    }
+    if (lane == 63) wsc[w] = inc;
+    __syncthreads();
+    uint32_t excl = inc - cnt;
+    for (int i = 0; i < w; ++i) excl += wsc[i];
+    uint64_t pos = base[blockIdx.x] + excl;
+#pragma unroll
+    for (int j = 0; j < IPT; ++j) {
+        if (keep[j]) {
+            out_k[pos] = kk[j];
+            out_v[pos] = vv[j];
+            pos++;
+        }
+    }
+}
+
+hipError_t narrow_map(hipStream_t s, const int64_t *in_k, const int64_t *in_v,
+                      uint64_t n, int op, int64_t p0, int64_t *out_k, int64_t *out_v) {
+    uint32_t nb = nblocks_for(n ? n : 1);
+    uint32_t gb = nb < 2048 ? (nb ? nb : 1) : 2048;
+    ProfScope ps("map", s);
+    hipLaunchKernelGGL(k_map_pairs, dim3(gb), dim3(BLOCK), 0, s, in_k, in_v, n, op, p0,
+                       out_k, out_v);
+    return hipGetLastError();
+}
+
+hipError_t narrow_filter(hipStream_t s, const int64_t *in_k, const int64_t *in_v,
+                         uint64_t n, int pred, int64_t p0, int64_t p1,
+                         int64_t *out_k, int64_t *out_v, uint64_t *h_nout, Ws &ws) {
+    if (n == 0) { *h_nout = 0; return hipSuccess; }
+    uint32_t nb = nblocks_for(n);
+    uint32_t *bc = (uint32_t *)ws.take(((size_t)nb + 1) * 4);
+    if (!bc) return hipErrorOutOfMemory;
+    {
+        ProfScope ps("filter_count", s);
+        hipLaunchKernelGGL(k_filter_count, dim3(nb), dim3(BLOCK), 0, s, in_k, in_v, n,
+                           pred, p0, p1, bc);
+        HIP_TRY(hipGetLastError());
+    }
+    HIP_TRY(hipMemsetAsync(bc + nb, 0, 4, s));
+    {
+        Ws w2 = ws;
+        HIP_TRY(scan_u32_excl(s, bc, (uint64_t)nb + 1, w2));
+    }
+    uint32_t total = 0;
+    HIP_TRY(hipMemcpyAsync(&total, bc + nb, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    {
+        ProfScope ps("filter_emit", s);
+        hipLaunchKernelGGL(k_filter_emit, dim3(nb), dim3(BLOCK), 0, s, in_k, in_v, n,
+                           pred, p0, p1, bc, out_k, out_v);
+        HIP_TRY(hipGetLastError());
+    }
+    *h_nout = total;
+    return hipSuccess;
+}
+
+/* ------------------------------------------------------------------ */
 /* misc host entries                                                   */
 
 hipError_t gen_uniform(hipStream_t s, int64_t *keys, int64_t *vals, uint64_t n,

@@ -24,6 +24,14 @@ OP_SUM_F64 = 2
 OP_MIN_I64 = 3
 OP_MAX_I64 = 4
 
+MAP_VALUES_ADD = 0
+MAP_VALUES_MUL = 1
+MAP_KEYS_ADD = 2
+MAP_SWAP = 3
+PRED_KEY_MOD_EQ = 0
+PRED_VAL_GT = 1
+PRED_KEY_IN_RANGE = 2
+
 _lib = None
 
 
@@ -144,6 +152,21 @@ class Rdd:
                "reduce_by_key", self.ctx._c)
         return Rdd(self.ctx, out.value,
                    np.float64 if op == OP_SUM_F64 else np.int64)
+
+    def map(self, op, p0=0):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_map(self.ctx._c, ctypes.c_uint64(self.h),
+                                  ctypes.c_int(op), ctypes.c_int64(p0),
+                                  ctypes.byref(out)), "map", self.ctx._c)
+        return Rdd(self.ctx, out.value, np.int64)
+
+    def filter(self, pred, p0=0, p1=0):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_filter(self.ctx._c, ctypes.c_uint64(self.h),
+                                     ctypes.c_int(pred), ctypes.c_int64(p0),
+                                     ctypes.c_int64(p1), ctypes.byref(out)),
+               "filter", self.ctx._c)
+        return Rdd(self.ctx, out.value, np.int64)
 
     def count_by_value(self, nparts=256):
         out = ctypes.c_uint64()

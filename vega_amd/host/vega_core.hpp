@@ -65,6 +65,17 @@ class PairRdd {
         check(vega_gpu_join(c_, h_, other.h_, num_splits, &out), "join", c_);
         return PairRdd(c_, out, false);
     }
+    /* rdd.rs:199-235 narrow transforms as op-enums (device-resident) */
+    PairRdd map(vega_map_op_t op, int64_t p0 = 0) const {
+        vega_rdd_t out = 0;
+        check(vega_gpu_map(c_, h_, op, p0, &out), "map", c_);
+        return PairRdd(c_, out, false);
+    }
+    PairRdd filter(vega_pred_t pred, int64_t p0 = 0, int64_t p1 = 0) const {
+        vega_rdd_t out = 0;
+        check(vega_gpu_filter(c_, h_, pred, p0, p1, &out), "filter", c_);
+        return PairRdd(c_, out, false);
+    }
     /* rdd.rs:449-459: counts over the VALUE column */
     PairRdd count_by_value(uint32_t num_splits = 256) const {
         vega_rdd_t out = 0;
