@@ -57,6 +57,7 @@ def test_device_gen_matches_host(ctx):
     (2, 1, 8),
     (100_000, 0, 9),      # all keys == 0 (zero active radix passes)
     (65536, 2, 10),       # 4 distinct keys
+    (500_000, 64, 11),    # full i64 keys incl. negatives
 ])
 def test_reduce_by_key_i64(ctx, n, bits, seed):
     from vega_amd import gpu
@@ -66,6 +67,17 @@ def test_reduce_by_key_i64(ctx, n, bits, seed):
     hk, hv = datagen.uniform_pairs(seed, n, key_bits=bits)
     ok, ov = oc.reduce_by_key_i64(hk, hv, 256, 256)
     assert sorted_pairs(gk, gv) == sorted_pairs(ok, ov)
+    rdd.free(); red.free()
+
+
+def test_reduce_wrapping_overflow_gpu(ctx):
+    from vega_amd import gpu
+    k = np.zeros(4, dtype=np.int64)
+    v = np.array([2**62] * 4, dtype=np.int64)
+    rdd = ctx.make_rdd(k, v)
+    red = rdd.reduce_by_key(gpu.OP_SUM_I64)
+    gk, gv = red.collect()
+    assert gk.tolist() == [0] and gv.tolist() == [0]
     rdd.free(); red.free()
 
 
@@ -191,6 +203,17 @@ def test_sort_by_key(ctx, n, bits, seed):
     assert (gk == ok).all(), "keys not in signed ascending stable order"
     assert (gv == ov).all(), "stability violated (values out of row order)"
     rdd.free(); srt.free()
+
+
+def test_group_by_key_full_values(ctx):
+    k, v = datagen.uniform_pairs(71, 50_000, key_bits=8)
+    rdd = ctx.make_rdd(k, v)
+    gk, off, gv = rdd.group_by_key()
+    import pyref
+    ref = pyref.group_by_key(k, v)
+    got = {int(gk[i]): gv[off[i]:off[i + 1]].tolist() for i in range(len(gk))}
+    assert got == ref  # per-group value ORDER = row order (stable sort)
+    rdd.free()
 
 
 # ---------------- golden vectors through the GPU ----------------

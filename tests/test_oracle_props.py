@@ -81,6 +81,14 @@ def test_reduce_f64_close_to_numpy():
         assert abs(got[int(key)] - ref) <= 1e-9 * max(1.0, abs(ref))
 
 
+def test_reduce_wrapping_overflow():
+    # Rust release-mode i64 add wraps; the engine and oracle must match that
+    k = np.zeros(4, dtype=np.int64)
+    v = np.array([2**62, 2**62, 2**62, 2**62], dtype=np.int64)
+    ok, ov = oc.reduce_by_key_i64(k, v, 2, 2)
+    assert ok.tolist() == [0] and ov.tolist() == [0]  # 4*2^62 wraps to 0
+
+
 def test_slice_bounds_reference_formula():
     # parallel_collection_rdd.rs:116-145: partition p = [p*n/P, (p+1)*n/P)
     for n, p in [(15, 4), (9, 4), (9, 2), (7, 2), (10, 3), (0, 4), (5, 7), (10**9, 256)]:

@@ -175,6 +175,18 @@ class Rdd:
                "count_by_value", self.ctx._c)
         return Rdd(self.ctx, out.value, np.int64)
 
+    def group_by_key(self):
+        """Full groups (pair_rdd.rs:35-52): (keys, offsets, values) with
+        values in row order per group (via the stable sort_by_key)."""
+        srt = self.sort_by_key()
+        k, v = srt.collect()
+        srt.free()
+        if len(k) == 0:
+            return k, np.zeros(1, dtype=np.int64), v
+        heads = np.flatnonzero(np.concatenate([[True], k[1:] != k[:-1]]))
+        offsets = np.concatenate([heads, [len(k)]]).astype(np.int64)
+        return k[heads], offsets, v
+
     def group_count(self, nparts=256):
         out = ctypes.c_uint64()
         _check(lib().vega_gpu_group_count(self.ctx._c, ctypes.c_uint64(self.h),
