@@ -173,12 +173,13 @@ int vega_dev_partition_range_i64(void *stream, const int64_t *keys, const int64_
                                  int64_t *out_k, int64_t *out_v, uint64_t *h_counts,
                                  void *d_ws, size_t ws_bytes);
 
-/* reduce-side sort+segmented-aggregate (replaces shuffled_rdd.rs:154-164's
- * HashMap merge_combiners): stable LSB radix sort of (k,v) by key then one
- * combiner per equal-key run. in_k/in_v are NOT modified. out arrays must
- * hold n rows (worst case all-distinct). *h_nout = #distinct keys (host,
- * after internal sync). vals/out_v are int64 for SUM_I64/COUNT/MIN/MAX,
- * double for SUM_F64. */
+/* reduce-side grouping + segmented aggregate (replaces
+ * shuffled_rdd.rs:154-164's HashMap merge_combiners): adaptive grouping sort
+ * (skipped key radix for narrow keys, else 32/40-bit splitmix64-hash radix
+ * with an exact collision cleanup) then one combiner per equal-key run.
+ * in_k/in_v are NOT modified. out arrays must hold n rows (worst case
+ * all-distinct). *h_nout = #distinct keys (host, after internal sync).
+ * vals/out_v are int64 for SUM_I64/COUNT/MIN/MAX, double for SUM_F64. */
 int vega_dev_sort_reduce(void *stream, const int64_t *in_k, const void *in_v,
                          uint64_t n, int op, int64_t *out_k, void *out_v,
                          uint64_t *h_nout, void *d_ws, size_t ws_bytes);

@@ -165,7 +165,9 @@ static int reduce_common(vega_ctx *c, vega_rdd_t rdd, int op, uint32_t nparts,
                          vega_rdd_t *out) {
     RddImpl *r = get_rdd(c, rdd);
     if (!r) return VEGA_ERR_INVALID;
-    if ((op == VEGA_OP_SUM_F64) != (r->vtype == 1)) return VEGA_ERR_INVALID;
+    /* SUM_F64 needs f64 values; COUNT ignores values; the i64 ops need i64 */
+    if (op == VEGA_OP_SUM_F64 && r->vtype != 1) return VEGA_ERR_INVALID;
+    if (op != VEGA_OP_SUM_F64 && op != VEGA_OP_COUNT && r->vtype == 1) return VEGA_ERR_INVALID;
     int rc = ensure_ws(c, r->n);
     if (rc) return rc;
     RddImpl *o;
