@@ -174,6 +174,17 @@ def test_group_count_zipf(ctx):
     rdd.free(); red.free()
 
 
+def test_group_count_on_f64_values(ctx):
+    # group_by_key->count must work for f64-valued RDDs (values ignored)
+    hk, hv = datagen.uniform_pairs_f64(23, 200_000, key_bits=8)
+    rdd = ctx.make_rdd(hk, hv)
+    red = rdd.group_count()
+    gk, gv = red.collect()
+    ok, ov = oc.group_count_i64(hk, np.zeros_like(hk), 8, 8)
+    assert sorted_pairs(gk, gv) == sorted_pairs(ok, ov)
+    rdd.free(); red.free()
+
+
 def test_count_by_value_golden(ctx):
     # test_pair_rdd.rs:85-109 via the dedicated API entry
     vals = np.array([1, 2, 1, 3, 2, 3, 3, 2, 3], dtype=np.int64)

@@ -130,3 +130,16 @@ def test_datagen_deterministic_and_sliceable():
     # Zipf skew: most-frequent key should dominate
     _, counts = np.unique(z, return_counts=True)
     assert counts.max() > 20000 * 0.05
+
+
+def test_oracle_fuzz_vs_pyref():
+    rng = np.random.RandomState(1234)
+    for _ in range(20):
+        n = int(rng.randint(1, 3000))
+        bits = int(rng.randint(1, 20))
+        pin = int(rng.randint(1, 17))
+        pout = int(rng.randint(1, 300))
+        seed = int(rng.randint(0, 1 << 30))
+        k, v = rand_pairs(seed, n, bits)
+        ok, ov = oc.reduce_by_key_i64(k, v, pin, pout)
+        assert dict(zip(ok.tolist(), ov.tolist())) == pyref.reduce_by_key(k, v)
