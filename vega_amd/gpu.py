@@ -168,6 +168,13 @@ class Rdd:
                "filter", self.ctx._c)
         return Rdd(self.ctx, out.value, np.int64)
 
+    def distinct(self, nparts=256):
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_distinct(self.ctx._c, ctypes.c_uint64(self.h),
+                                       ctypes.c_uint32(nparts), ctypes.byref(out)),
+               "distinct", self.ctx._c)
+        return Rdd(self.ctx, out.value, np.int64)
+
     def count_by_value(self, nparts=256):
         out = ctypes.c_uint64()
         _check(lib().vega_gpu_count_by_value(self.ctx._c, ctypes.c_uint64(self.h),
