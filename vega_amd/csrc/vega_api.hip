@@ -245,23 +245,42 @@ int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
 }
 
 /* inner join (pair_rdd.rs:104-121 via cogroup co_grouped_rdd.rs:206-249):
- * sort both sides, then sort-merge count + emit (K4). */
+ * bring both sides into the GROUPING order (4-5 hash passes instead of the
+ * full 8-pass signed sort) and sort-merge with the (h32,key) comparator. */
 int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
                   vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
     RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
     if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
-    vega_rdd_t ha = 0, hb = 0;
-    int rc = vega_gpu_sort_by_key(c, a, &ha);
+    uint64_t nmax = ra->n > rb->n ? ra->n : rb->n;
+    int rc = ensure_ws(c, nmax);
     if (rc) return rc;
-    rc = vega_gpu_sort_by_key(c, b, &hb);
+    /* grouped copies of both sides */
+    vega_rdd_t ha = 0, hb = 0;
+    RddImpl *sa, *sb;
+    rc = new_rdd(c, ra->n ? ra->n : 1, 0, nparts, &sa, &ha);
+    if (rc) return rc;
+    rc = new_rdd(c, rb->n ? rb->n : 1, 0, nparts, &sb, &hb);
     if (rc) { vega_gpu_free_rdd(c, ha); return rc; }
-    RddImpl *sa = get_rdd(c, ha), *sb = get_rdd(c, hb);
+    sa->n = ra->n;
+    sb->n = rb->n;
+    if (ra->n) {
+        CTX_TRY(c, hipMemcpyAsync(sa->d_k, ra->d_k, ra->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(sa->d_v, ra->d_v, ra->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        Ws wsa(c->ws, c->ws_bytes);
+        CTX_TRY(c, group_pairs_inplace(c->stream, sa->d_k, (int64_t *)sa->d_v, sa->n, wsa));
+    }
+    if (rb->n) {
+        CTX_TRY(c, hipMemcpyAsync(sb->d_k, rb->d_k, rb->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(sb->d_v, rb->d_v, rb->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        Ws wsb(c->ws, c->ws_bytes);
+        CTX_TRY(c, group_pairs_inplace(c->stream, sb->d_k, (int64_t *)sb->d_v, sb->n, wsb));
+    }
     uint64_t total = 0;
     {
         Ws ws(c->ws, c->ws_bytes);
         hipError_t e = join_sorted(c->stream, sa->d_k, (const int64_t *)sa->d_v, sa->n,
-                                   sb->d_k, (const int64_t *)sb->d_v, sb->n,
+                                   sb->d_k, (const int64_t *)sb->d_v, sb->n, 1,
                                    nullptr, nullptr, nullptr, 0, &total, ws);
         if (e != hipSuccess) {
             vega_gpu_free_rdd(c, ha); vega_gpu_free_rdd(c, hb);
@@ -278,7 +297,7 @@ int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
         Ws ws(c->ws, c->ws_bytes);
         uint64_t n2 = 0;
         hipError_t e = join_sorted(c->stream, sa->d_k, (const int64_t *)sa->d_v, sa->n,
-                                   sb->d_k, (const int64_t *)sb->d_v, sb->n,
+                                   sb->d_k, (const int64_t *)sb->d_v, sb->n, 1,
                                    o->d_k, (int64_t *)o->d_v, (int64_t *)o->d_v2,
                                    total, &n2, ws);
         vega_gpu_free_rdd(c, ha);
@@ -464,9 +483,29 @@ int vega_dev_join_sorted(void *stream, const int64_t *ak, const int64_t *av, uin
                          int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                          uint64_t cap, uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
     Ws ws(d_ws, ws_bytes);
-    hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb,
+    hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb, 0,
                                out_k, out_va, out_vb, cap, h_nout, ws);
     if (e == hipErrorNotSupported) return VEGA_ERR_UNSUPPORTED;
+    return e == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
+}
+
+/* bring rows into the GROUPING order in place ((h32,key) lexicographic;
+ * the cheap order vega_dev_join_grouped expects) */
+int vega_dev_group_pairs_i64(void *stream, int64_t *keys, int64_t *vals, uint64_t n,
+                             void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    hipError_t e = group_pairs_inplace((hipStream_t)stream, keys, vals, n, ws);
+    return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
+}
+
+/* sort-merge inner join of two sides in the GROUPING order */
+int vega_dev_join_grouped(void *stream, const int64_t *ak, const int64_t *av, uint64_t na,
+                          const int64_t *bk, const int64_t *bv, uint64_t nb,
+                          int64_t *out_k, int64_t *out_va, int64_t *out_vb,
+                          uint64_t cap, uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb, 1,
+                               out_k, out_va, out_vb, cap, h_nout, ws);
     return e == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
 }
 

@@ -88,11 +88,17 @@ hipError_t narrow_filter(hipStream_t s, const int64_t *in_k, const int64_t *in_v
 hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
                           uint64_t n, uint64_t *h_sum, Ws &ws);
 
-/* sort-merge inner join of key-sorted sides */
+/* sort-merge inner join; hash_order=0: signed-key-sorted sides, 1: sides in
+ * the grouping order ((h32,key) lexicographic, from group_pairs_inplace) */
 hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint64_t na,
                        const int64_t *bk, const int64_t *bv, uint64_t nb,
+                       int hash_order,
                        int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                        uint64_t cap, uint64_t *h_nout, Ws &ws);
+
+/* in-place grouping-order sort (the cheap 4-5 pass order joins use) */
+hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
+                               uint64_t n, Ws &ws);
 
 size_t ws_bytes_for(uint64_t n);
 

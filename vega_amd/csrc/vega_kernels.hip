@@ -1464,6 +1464,17 @@ hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
     return hipStreamSynchronize(s);
 }
 
+hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
+                               uint64_t n, Ws &ws) {
+    const uint64_t *rk, *rv;
+    HIP_TRY(group_sort_u64(s, (const uint64_t *)keys, (const uint64_t *)vals, n, ws, &rk, &rv));
+    if ((const uint64_t *)keys != rk) {
+        HIP_TRY(hipMemcpyAsync(keys, rk, n * 8, hipMemcpyDeviceToDevice, s));
+        HIP_TRY(hipMemcpyAsync(vals, rv, n * 8, hipMemcpyDeviceToDevice, s));
+    }
+    return hipSuccess;
+}
+
 size_t ws_bytes_for(uint64_t n) {
     uint64_t nb = nblocks_for(n ? n : 1);
     size_t b = 0;
@@ -1487,8 +1498,19 @@ size_t ws_bytes_for(uint64_t n) {
  * equal-key run in B (log2 nb probes; upper tree levels stay in L2/L3),
  * count + base, exclusive scan, then emit the cross product. */
 
+/* the grouping order is (h32(k), k) unsigned-lexicographic: the hash radix
+ * sorts by h32 and the cleanup key-sorts within equal-h32 runs */
+__device__ __forceinline__ bool lex_less_grouped(int64_t a, int64_t b) {
+    uint32_t ha = (uint32_t)vega_hash_u64((uint64_t)a);
+    uint32_t hb = (uint32_t)vega_hash_u64((uint64_t)b);
+    if (ha != hb) return ha < hb;
+    return (uint64_t)a < (uint64_t)b;
+}
+
+/* hash_order=0: sides are signed-key sorted; 1: sides are in grouping order */
 __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
-                             uint64_t nb, uint32_t *counts, uint32_t *b_lo) {
+                             uint64_t nb, int hash_order, uint32_t *counts,
+                             uint32_t *b_lo) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < na; i += stride) {
         int64_t k = ak[i];
@@ -1496,14 +1518,16 @@ __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
         uint64_t lo = 0, hi = nb;
         while (lo < hi) {
             uint64_t m = (lo + hi) >> 1;
-            if (bk[m] < k) lo = m + 1; else hi = m;
+            bool less = hash_order ? lex_less_grouped(bk[m], k) : (bk[m] < k);
+            if (less) lo = m + 1; else hi = m;
         }
         uint64_t lb = lo;
         /* upper bound */
         hi = nb;
         while (lo < hi) {
             uint64_t m = (lo + hi) >> 1;
-            if (bk[m] <= k) lo = m + 1; else hi = m;
+            bool gt = hash_order ? lex_less_grouped(k, bk[m]) : (k < bk[m]);
+            if (!gt) lo = m + 1; else hi = m;
         }
         counts[i] = (uint32_t)(lo - lb);
         b_lo[i] = (uint32_t)lb;
@@ -1532,6 +1556,7 @@ __global__ void k_join_emit(const int64_t *ak, const int64_t *av, uint64_t na,
 
 hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint64_t na,
                        const int64_t *bk, const int64_t *bv, uint64_t nb,
+                       int hash_order,
                        int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                        uint64_t cap, uint64_t *h_nout, Ws &ws) {
     if (na == 0 || nb == 0) { *h_nout = 0; return hipSuccess; }
@@ -1542,7 +1567,8 @@ hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint
     uint32_t gb = nb_grid < 2048 ? nb_grid : 2048;
     {
         ProfScope ps("join_count", s);
-        hipLaunchKernelGGL(k_join_count, dim3(gb), dim3(BLOCK), 0, s, ak, na, bk, nb, counts, b_lo);
+        hipLaunchKernelGGL(k_join_count, dim3(gb), dim3(BLOCK), 0, s, ak, na, bk, nb,
+                           hash_order, counts, b_lo);
         HIP_TRY(hipGetLastError());
     }
     HIP_TRY(hipMemsetAsync(counts + na, 0, 4, s));
