@@ -471,7 +471,7 @@ template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
 __global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
-    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, DF df) {
+    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, int *d_abort, DF df) {
     /* 512 threads = 8 waves per block (16 waves/CU at 2 blocks): each wave
      * ranks a 512-row chunk of the 4096-row tile. IN_PK/OUT_PK: interleaved
      * (k,v) rows — one 16-B vector access per row. */
@@ -589,7 +589,11 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         if (vb > 0) {
             gdesc_t *col = (gdesc_t *)(desc + t);
             int64_t j = (int64_t)vb - 1;
+            uint32_t spins = 0;
             while (j >= 0) {
+                /* bounded spin: a lost predecessor can never wedge the GPU —
+                 * set the abort flag, bail, and let the host fail loudly */
+                if (++spins > (1u << 26)) { *d_abort = 1; break; }
                 /* probe up to 4 predecessors with independent loads */
                 unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
                 int navail = (j >= 3) ? 4 : (int)(j + 1);
@@ -663,7 +667,7 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    uint64_t n, const uint32_t *gbase_d,
                                    unsigned long long *desc, uint32_t *ticket,
                                    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out,
-                                   bool has_vals, bool in_pk, bool out_pk,
+                                   int *d_abort, bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
@@ -672,19 +676,19 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
         hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
     } else if (!in_pk && !out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, df);
     } else if (!in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
     } else if (in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
     } else {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, df);
     }
     return hipGetLastError();
 }
@@ -765,8 +769,10 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *gbase_d = (uint32_t *)ws.take(8 * 256 * 4);
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
-    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8)
+    int *d_abort = (int *)ws.take(256);
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort)
         return hipErrorOutOfMemory;
+    HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
 
     /* exact per-byte histograms: pass skipping + the onesweep global bases */
     HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
@@ -812,14 +818,18 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         if (p == 7 && signed_order) {
             RadixDigitTopSigned df{56};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, has_vals, in_pk, out_pk, df, "radix_scatter"));
         } else {
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, has_vals, in_pk, out_pk, df, "radix_scatter"));
         }
         cur = dk;
     }
+    int ab = 0;
+    HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (ab) return hipErrorUnknown; /* lookback bailed: fail loudly */
     *res_k = cur;
     *res_v = has_vals ? cur + n : nullptr;
     return hipSuccess;
@@ -904,9 +914,11 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
+    int *d_abort = (int *)ws.take(256);
     uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
-    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !h32buf)
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !d_abort || !h32buf)
         return hipErrorOutOfMemory;
+    HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
 
     static thread_local uint32_t hh[8 * 256];
     static thread_local uint32_t gb_host[8 * 256];
@@ -958,7 +970,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur, i == 0 ? in_v : nullptr,
                                      n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, true, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, true, in_pk, out_pk, df, "radix_scatter"));
             cur = dk;
         }
         *rk = cur;
@@ -976,7 +988,10 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(exact_hists(false, &active));
         if (active <= 5) { /* narrow keys: skipped key sort groups exactly */
             HIP_TRY(run_key_passes(res_k, res_v));
-            return hipSuccess;
+            int ab = 0;
+            HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
+            HIP_TRY(hipStreamSynchronize(s));
+            return ab ? hipErrorUnknown : hipSuccess;
         }
         /* sample lied; fall through to the hash path */
     }
@@ -996,7 +1011,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur_k, i == 0 ? in_v : nullptr,
                                  n, gbase_d + i * 256, desc, ticket,
                                  dk, dv, out_pk ? nullptr : h32buf,
-                                 true, in_pk, out_pk, df, "radix_scatter"));
+                                 d_abort, true, in_pk, out_pk, df, "radix_scatter"));
         cur_k = dk;
         cur_v = out_pk ? nullptr : dk + n;
     }
@@ -1016,6 +1031,10 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(exact_hists(false, &active));
         HIP_TRY(run_key_passes(&cur_k, &cur_v));
     }
+    int ab = 0;
+    HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (ab) return hipErrorUnknown; /* lookback bailed: fail loudly */
     *res_k = cur_k;
     *res_v = cur_v;
     return hipSuccess;
