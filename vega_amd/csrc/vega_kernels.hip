@@ -899,11 +899,17 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
  * Adaptive: if the per-byte key histograms show <= 5 active radix passes,
  * the plain key sort is cheaper; else 5 hash-byte passes + cleanup (with a
  * full-key-sort fallback on oversized dirty runs). */
+/* order_tag (optional out): 0 = result is FULL-KEY unsigned-ascending
+ * (narrow-key path, fallback, or trivial); 4 = result is (h32(key), key)
+ * unsigned-lexicographic (4 hash-byte passes + h32 cleanup). force_hbytes=4
+ * pins the hash order regardless of n (joins need a stable comparator);
+ * 0 = adaptive (reduce path). */
 hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
-                          uint64_t n, Ws &ws,
+                          uint64_t n, int force_hbytes, int *order_tag, Ws &ws,
                           const uint64_t **res_k, const uint64_t **res_v) {
     *res_k = in_k;
     *res_v = in_v;
+    if (order_tag) *order_tag = 0;
     if (n <= 1) return hipSuccess;
     uint32_t nb = nblocks_for(n);
 
@@ -998,8 +1004,9 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 
     /* hash grouping: enough hash bytes to keep the expected bucket load
      * factor <= ~0.25 (4 bytes up to 2^30 rows, 5 above) — fewer passes,
-     * slightly busier cleanup */
-    const int hbytes = (n <= (1ULL << 30)) ? 4 : 5;
+     * slightly busier cleanup. force_hbytes pins the byte count (and with
+     * it the order contract) for joins. */
+    const int hbytes = force_hbytes ? force_hbytes : ((n <= (1ULL << 30)) ? 4 : 5);
     int active5 = 0;
     HIP_TRY(exact_hists(true, &active5));
     for (int i = 0; i < hbytes; ++i) {
@@ -1035,6 +1042,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
     if (ab) return hipErrorUnknown; /* lookback bailed: fail loudly */
+    if (order_tag && !err) *order_tag = (hbytes == 4) ? 4 : 0;
     *res_k = cur_k;
     *res_v = cur_v;
     return hipSuccess;
@@ -1484,9 +1492,10 @@ hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
 }
 
 hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
-                               uint64_t n, Ws &ws) {
+                               uint64_t n, int *order_tag, Ws &ws) {
     const uint64_t *rk, *rv;
-    HIP_TRY(group_sort_u64(s, (const uint64_t *)keys, (const uint64_t *)vals, n, ws, &rk, &rv));
+    HIP_TRY(group_sort_u64(s, (const uint64_t *)keys, (const uint64_t *)vals, n,
+                           /*force_hbytes=*/4, order_tag, ws, &rk, &rv));
     if ((const uint64_t *)keys != rk) {
         HIP_TRY(hipMemcpyAsync(keys, rk, n * 8, hipMemcpyDeviceToDevice, s));
         HIP_TRY(hipMemcpyAsync(vals, rv, n * 8, hipMemcpyDeviceToDevice, s));
@@ -1517,18 +1526,21 @@ size_t ws_bytes_for(uint64_t n) {
  * equal-key run in B (log2 nb probes; upper tree levels stay in L2/L3),
  * count + base, exclusive scan, then emit the cross product. */
 
-/* the grouping order is (h32(k), k) unsigned-lexicographic: the hash radix
- * sorts by h32 and the cleanup key-sorts within equal-h32 runs */
-__device__ __forceinline__ bool lex_less_grouped(int64_t a, int64_t b) {
+/* join comparator modes (must match how the sides were sorted):
+ *   0: signed key ascending (sort_by_key order; external callers)
+ *   1: unsigned key ascending (radix_sort_u64 signed_order=false)
+ *   2: (h32(key), key) unsigned-lexicographic (group order, tag 4) */
+__device__ __forceinline__ bool join_less(int mode, int64_t a, int64_t b) {
+    if (mode == 0) return a < b;
+    if (mode == 1) return (uint64_t)a < (uint64_t)b;
     uint32_t ha = (uint32_t)vega_hash_u64((uint64_t)a);
     uint32_t hb = (uint32_t)vega_hash_u64((uint64_t)b);
     if (ha != hb) return ha < hb;
     return (uint64_t)a < (uint64_t)b;
 }
 
-/* hash_order=0: sides are signed-key sorted; 1: sides are in grouping order */
 __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
-                             uint64_t nb, int hash_order, uint32_t *counts,
+                             uint64_t nb, int mode, uint32_t *counts,
                              uint32_t *b_lo) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < na; i += stride) {
@@ -1537,16 +1549,14 @@ __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
         uint64_t lo = 0, hi = nb;
         while (lo < hi) {
             uint64_t m = (lo + hi) >> 1;
-            bool less = hash_order ? lex_less_grouped(bk[m], k) : (bk[m] < k);
-            if (less) lo = m + 1; else hi = m;
+            if (join_less(mode, bk[m], k)) lo = m + 1; else hi = m;
         }
         uint64_t lb = lo;
         /* upper bound */
         hi = nb;
         while (lo < hi) {
             uint64_t m = (lo + hi) >> 1;
-            bool gt = hash_order ? lex_less_grouped(k, bk[m]) : (k < bk[m]);
-            if (!gt) lo = m + 1; else hi = m;
+            if (!join_less(mode, k, bk[m])) lo = m + 1; else hi = m;
         }
         counts[i] = (uint32_t)(lo - lb);
         b_lo[i] = (uint32_t)lb;
@@ -1575,7 +1585,7 @@ __global__ void k_join_emit(const int64_t *ak, const int64_t *av, uint64_t na,
 
 hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint64_t na,
                        const int64_t *bk, const int64_t *bv, uint64_t nb,
-                       int hash_order,
+                       int mode,
                        int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                        uint64_t cap, uint64_t *h_nout, Ws &ws) {
     if (na == 0 || nb == 0) { *h_nout = 0; return hipSuccess; }
@@ -1587,7 +1597,7 @@ hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint
     {
         ProfScope ps("join_count", s);
         hipLaunchKernelGGL(k_join_count, dim3(gb), dim3(BLOCK), 0, s, ak, na, bk, nb,
-                           hash_order, counts, b_lo);
+                           mode, counts, b_lo);
         HIP_TRY(hipGetLastError());
     }
     HIP_TRY(hipMemsetAsync(counts + na, 0, 4, s));
