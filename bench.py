@@ -215,18 +215,29 @@ def main():
         if world == 1:
             out_k[:rows].copy_(k); out_v[:rows].copy_(v)
             sb_k[:rows].copy_(kb); sb_v[:rows].copy_(vb)
-            gpu.dev_sort_pairs(out_k[:rows], out_v[:rows], ws)
-            gpu.dev_sort_pairs(sb_k[:rows], sb_v[:rows], ws)
-            nout = gpu.dev_join_sorted(out_k[:rows], out_v[:rows],
-                                       sb_k[:rows], sb_v[:rows], jk, jva, jvb, ws)
+            ta = gpu.dev_group_pairs(out_k[:rows], out_v[:rows], ws)
+            tb = gpu.dev_group_pairs(sb_k[:rows], sb_v[:rows], ws)
+            if ta == 4 and tb == 4:
+                nout = gpu.dev_join_grouped(out_k[:rows], out_v[:rows],
+                                            sb_k[:rows], sb_v[:rows], 2, jk, jva, jvb, ws)
+            else:  # harmonize to the unsigned-key order
+                gpu.dev_sort_pairs(out_k[:rows], out_v[:rows], ws)
+                gpu.dev_sort_pairs(sb_k[:rows], sb_v[:rows], ws)
+                nout = gpu.dev_join_sorted(out_k[:rows], out_v[:rows],
+                                           sb_k[:rows], sb_v[:rows], jk, jva, jvb, ws)
         else:
             ca = gpu.dev_partition(k, v, world, pk, pv, ws)
             rak, rav = shuffle.all_to_all_kv(pk, pv, ca.astype(np.int64).tolist())
             cb = gpu.dev_partition(kb, vb, world, pk, pv, ws)
             rbk, rbv = shuffle.all_to_all_kv(pk, pv, cb.astype(np.int64).tolist())
-            gpu.dev_sort_pairs(rak, rav, ws)
-            gpu.dev_sort_pairs(rbk, rbv, ws)
-            nout = gpu.dev_join_sorted(rak, rav, rbk, rbv, jk, jva, jvb, ws)
+            ta = gpu.dev_group_pairs(rak, rav, ws)
+            tb = gpu.dev_group_pairs(rbk, rbv, ws)
+            if ta == 4 and tb == 4:
+                nout = gpu.dev_join_grouped(rak, rav, rbk, rbv, 2, jk, jva, jvb, ws)
+            else:
+                gpu.dev_sort_pairs(rak, rav, ws)
+                gpu.dev_sort_pairs(rbk, rbv, ws)
+                nout = gpu.dev_join_sorted(rak, rav, rbk, rbv, jk, jva, jvb, ws)
 
     step = {"reduce": lambda: step_reduce(
                 gpu.OP_SUM_F64 if args.dtype == "f64" else gpu.OP_SUM_I64),

@@ -198,6 +198,20 @@ int vega_dev_join_sorted(void *stream,
                          uint64_t cap, uint64_t *h_nout,
                          void *d_ws, size_t ws_bytes);
 
+/* bring rows into the GROUPING order in place ((h32(key), key) unsigned
+ * lexicographic — 4-5 hash radix passes + collision cleanup; cheaper than a
+ * full signed sort when only co-grouping matters, e.g. before a join) */
+int vega_dev_group_pairs_i64(void *stream, int64_t *keys, int64_t *vals, uint64_t n,
+                             int *h_order_tag, void *d_ws, size_t ws_bytes);
+
+/* sort-merge inner join; order_mode must match how BOTH sides were sorted:
+ * 0 = signed-key order, 1 = unsigned-key order, 2 = grouping order (tag 4) */
+int vega_dev_join_grouped(void *stream, const int64_t *ak, const int64_t *av, uint64_t na,
+                          const int64_t *bk, const int64_t *bv, uint64_t nb,
+                          int order_mode,
+                          int64_t *out_k, int64_t *out_va, int64_t *out_vb,
+                          uint64_t cap, uint64_t *h_nout, void *d_ws, size_t ws_bytes);
+
 /* order-independent multiset checksum of device rows (same formula as
  * oracle_checksum_pairs_i64) — large-size parity checks without D2H */
 int vega_dev_checksum_pairs(void *stream, const int64_t *keys, const int64_t *vals,

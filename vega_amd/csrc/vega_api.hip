@@ -176,7 +176,7 @@ static int reduce_common(vega_ctx *c, vega_rdd_t rdd, int op, uint32_t nparts,
     Ws ws(c->ws, c->ws_bytes);
     const uint64_t *sk, *sv;
     CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
-                              r->n, ws, &sk, &sv));
+                              r->n, 0, nullptr, ws, &sk, &sv));
     uint64_t nout = 0;
     CTX_TRY(c, seg_reduce(c->stream, sk, sv, r->n, op, (uint64_t *)o->d_k, o->d_v, &nout, ws));
     o->n = nout;
@@ -216,7 +216,7 @@ int vega_gpu_count_by_value(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts,
     Ws ws(c->ws, c->ws_bytes);
     const uint64_t *sk, *sv;
     CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_v, (const uint64_t *)r->d_k,
-                              r->n, ws, &sk, &sv));
+                              r->n, 0, nullptr, ws, &sk, &sv));
     uint64_t nout = 0;
     CTX_TRY(c, seg_reduce(c->stream, sk, sv, r->n, VEGA_OP_COUNT,
                           (uint64_t *)o->d_k, o->d_v, &nout, ws));
@@ -245,23 +245,69 @@ int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
 }
 
 /* inner join (pair_rdd.rs:104-121 via cogroup co_grouped_rdd.rs:206-249):
- * sort both sides, then sort-merge count + emit (K4). */
+ * bring both sides into the GROUPING order (4-5 hash passes instead of the
+ * full 8-pass signed sort) and sort-merge with the (h32,key) comparator. */
 int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
                   vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
     RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
     if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
-    vega_rdd_t ha = 0, hb = 0;
-    int rc = vega_gpu_sort_by_key(c, a, &ha);
+    uint64_t nmax = ra->n > rb->n ? ra->n : rb->n;
+    int rc = ensure_ws(c, nmax);
     if (rc) return rc;
-    rc = vega_gpu_sort_by_key(c, b, &hb);
+    /* grouped copies of both sides */
+    vega_rdd_t ha = 0, hb = 0;
+    RddImpl *sa, *sb;
+    rc = new_rdd(c, ra->n ? ra->n : 1, 0, nparts, &sa, &ha);
+    if (rc) return rc;
+    rc = new_rdd(c, rb->n ? rb->n : 1, 0, nparts, &sb, &hb);
     if (rc) { vega_gpu_free_rdd(c, ha); return rc; }
-    RddImpl *sa = get_rdd(c, ha), *sb = get_rdd(c, hb);
+    sa->n = ra->n;
+    sb->n = rb->n;
+    int tag_a = 0, tag_b = 0;
+    if (ra->n) {
+        CTX_TRY(c, hipMemcpyAsync(sa->d_k, ra->d_k, ra->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(sa->d_v, ra->d_v, ra->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        Ws wsa(c->ws, c->ws_bytes);
+        CTX_TRY(c, group_pairs_inplace(c->stream, sa->d_k, (int64_t *)sa->d_v, sa->n, &tag_a, wsa));
+    }
+    if (rb->n) {
+        CTX_TRY(c, hipMemcpyAsync(sb->d_k, rb->d_k, rb->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(sb->d_v, rb->d_v, rb->n * 8, hipMemcpyDeviceToDevice, c->stream));
+        Ws wsb(c->ws, c->ws_bytes);
+        CTX_TRY(c, group_pairs_inplace(c->stream, sb->d_k, (int64_t *)sb->d_v, sb->n, &tag_b, wsb));
+    }
+    /* both sides must share the comparator's order: if either side is not
+     * in the (h32,key) order (narrow keys or fallback), harmonize both to
+     * the full unsigned-key order */
+    int join_mode = 2;
+    if (tag_a != 4 || tag_b != 4) {
+        join_mode = 1;
+        const uint64_t *rk, *rv;
+        if (sa->n && tag_a == 4) {
+            Ws wsa(c->ws, c->ws_bytes);
+            CTX_TRY(c, radix_sort_u64(c->stream, (const uint64_t *)sa->d_k,
+                                      (const uint64_t *)sa->d_v, sa->n, true, false, wsa, &rk, &rv));
+            if ((const uint64_t *)sa->d_k != rk) {
+                CTX_TRY(c, hipMemcpyAsync(sa->d_k, rk, sa->n * 8, hipMemcpyDeviceToDevice, c->stream));
+                CTX_TRY(c, hipMemcpyAsync(sa->d_v, rv, sa->n * 8, hipMemcpyDeviceToDevice, c->stream));
+            }
+        }
+        if (sb->n && tag_b == 4) {
+            Ws wsb(c->ws, c->ws_bytes);
+            CTX_TRY(c, radix_sort_u64(c->stream, (const uint64_t *)sb->d_k,
+                                      (const uint64_t *)sb->d_v, sb->n, true, false, wsb, &rk, &rv));
+            if ((const uint64_t *)sb->d_k != rk) {
+                CTX_TRY(c, hipMemcpyAsync(sb->d_k, rk, sb->n * 8, hipMemcpyDeviceToDevice, c->stream));
+                CTX_TRY(c, hipMemcpyAsync(sb->d_v, rv, sb->n * 8, hipMemcpyDeviceToDevice, c->stream));
+            }
+        }
+    }
     uint64_t total = 0;
     {
         Ws ws(c->ws, c->ws_bytes);
         hipError_t e = join_sorted(c->stream, sa->d_k, (const int64_t *)sa->d_v, sa->n,
-                                   sb->d_k, (const int64_t *)sb->d_v, sb->n,
+                                   sb->d_k, (const int64_t *)sb->d_v, sb->n, join_mode,
                                    nullptr, nullptr, nullptr, 0, &total, ws);
         if (e != hipSuccess) {
             vega_gpu_free_rdd(c, ha); vega_gpu_free_rdd(c, hb);
@@ -278,7 +324,7 @@ int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
         Ws ws(c->ws, c->ws_bytes);
         uint64_t n2 = 0;
         hipError_t e = join_sorted(c->stream, sa->d_k, (const int64_t *)sa->d_v, sa->n,
-                                   sb->d_k, (const int64_t *)sb->d_v, sb->n,
+                                   sb->d_k, (const int64_t *)sb->d_v, sb->n, join_mode,
                                    o->d_k, (int64_t *)o->d_v, (int64_t *)o->d_v2,
                                    total, &n2, ws);
         vega_gpu_free_rdd(c, ha);
@@ -437,7 +483,7 @@ int vega_dev_sort_reduce(void *stream, const int64_t *in_k, const void *in_v,
     Ws ws(d_ws, ws_bytes);
     const uint64_t *sk, *sv;
     hipError_t e = group_sort_u64((hipStream_t)stream, (const uint64_t *)in_k,
-                                  (const uint64_t *)in_v, n, ws, &sk, &sv);
+                                  (const uint64_t *)in_v, n, 0, nullptr, ws, &sk, &sv);
     if (e != hipSuccess) return e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP;
     e = seg_reduce((hipStream_t)stream, sk, sv, n, op, (uint64_t *)out_k, out_v, h_nout, ws);
     return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
@@ -464,9 +510,31 @@ int vega_dev_join_sorted(void *stream, const int64_t *ak, const int64_t *av, uin
                          int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                          uint64_t cap, uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
     Ws ws(d_ws, ws_bytes);
-    hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb,
+    hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb, 0,
                                out_k, out_va, out_vb, cap, h_nout, ws);
     if (e == hipErrorNotSupported) return VEGA_ERR_UNSUPPORTED;
+    return e == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
+}
+
+/* bring rows into the GROUPING order in place ((h32,key) lexicographic;
+ * the cheap order vega_dev_join_grouped expects) */
+int vega_dev_group_pairs_i64(void *stream, int64_t *keys, int64_t *vals, uint64_t n,
+                             int *h_order_tag, void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    hipError_t e = group_pairs_inplace((hipStream_t)stream, keys, vals, n, h_order_tag, ws);
+    return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
+}
+
+/* sort-merge inner join; order_mode must match how BOTH sides are sorted:
+ * 0 signed-key, 1 unsigned-key, 2 = (h32,key) lex (group tag 4) */
+int vega_dev_join_grouped(void *stream, const int64_t *ak, const int64_t *av, uint64_t na,
+                          const int64_t *bk, const int64_t *bv, uint64_t nb,
+                          int order_mode,
+                          int64_t *out_k, int64_t *out_va, int64_t *out_vb,
+                          uint64_t cap, uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
+    Ws ws(d_ws, ws_bytes);
+    hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb, order_mode,
+                               out_k, out_va, out_vb, cap, h_nout, ws);
     return e == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
 }
 

@@ -57,8 +57,10 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 
 /* group equal keys adjacently for the reduce path (no total-order contract):
  * adaptive skipped key sort vs 40-bit hash sort + collision-run cleanup */
+/* order_tag out: 0 = full-key unsigned order, 4 = (h32,key) lex order.
+ * force_hbytes: 0 adaptive (reduce), 4 pinned order (joins). */
 hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
-                          uint64_t n, Ws &ws,
+                          uint64_t n, int force_hbytes, int *order_tag, Ws &ws,
                           const uint64_t **res_k, const uint64_t **res_v);
 
 /* segmented aggregate over key-sorted rows: one output row per equal-key
@@ -88,11 +90,18 @@ hipError_t narrow_filter(hipStream_t s, const int64_t *in_k, const int64_t *in_v
 hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
                           uint64_t n, uint64_t *h_sum, Ws &ws);
 
-/* sort-merge inner join of key-sorted sides */
+/* sort-merge inner join; hash_order=0: signed-key-sorted sides, 1: sides in
+ * the grouping order ((h32,key) lexicographic, from group_pairs_inplace) */
+/* mode: 0 signed-key order, 1 unsigned-key order, 2 (h32,key) lex order */
 hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint64_t na,
                        const int64_t *bk, const int64_t *bv, uint64_t nb,
+                       int mode,
                        int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                        uint64_t cap, uint64_t *h_nout, Ws &ws);
+
+/* in-place grouping-order sort (the cheap 4-5 pass order joins use) */
+hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
+                               uint64_t n, int *order_tag, Ws &ws);
 
 size_t ws_bytes_for(uint64_t n);
 

@@ -471,7 +471,7 @@ template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
 __global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
-    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, DF df) {
+    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, int *d_abort, DF df) {
     /* 512 threads = 8 waves per block (16 waves/CU at 2 blocks): each wave
      * ranks a 512-row chunk of the 4096-row tile. IN_PK/OUT_PK: interleaved
      * (k,v) rows — one 16-B vector access per row. */
@@ -589,7 +589,11 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         if (vb > 0) {
             gdesc_t *col = (gdesc_t *)(desc + t);
             int64_t j = (int64_t)vb - 1;
+            uint32_t spins = 0;
             while (j >= 0) {
+                /* bounded spin: a lost predecessor can never wedge the GPU —
+                 * set the abort flag, bail, and let the host fail loudly */
+                if (++spins > (1u << 26)) { *d_abort = 1; break; }
                 /* probe up to 4 predecessors with independent loads */
                 unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
                 int navail = (j >= 3) ? 4 : (int)(j + 1);
@@ -663,7 +667,7 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    uint64_t n, const uint32_t *gbase_d,
                                    unsigned long long *desc, uint32_t *ticket,
                                    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out,
-                                   bool has_vals, bool in_pk, bool out_pk,
+                                   int *d_abort, bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
@@ -672,19 +676,19 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
         hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
     } else if (!in_pk && !out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, df);
     } else if (!in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
     } else if (in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
     } else {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, df);
     }
     return hipGetLastError();
 }
@@ -765,8 +769,10 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *gbase_d = (uint32_t *)ws.take(8 * 256 * 4);
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
-    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8)
+    int *d_abort = (int *)ws.take(256);
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort)
         return hipErrorOutOfMemory;
+    HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
 
     /* exact per-byte histograms: pass skipping + the onesweep global bases */
     HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
@@ -812,14 +818,18 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         if (p == 7 && signed_order) {
             RadixDigitTopSigned df{56};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, has_vals, in_pk, out_pk, df, "radix_scatter"));
         } else {
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, has_vals, in_pk, out_pk, df, "radix_scatter"));
         }
         cur = dk;
     }
+    int ab = 0;
+    HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (ab) return hipErrorUnknown; /* lookback bailed: fail loudly */
     *res_k = cur;
     *res_v = has_vals ? cur + n : nullptr;
     return hipSuccess;
@@ -889,11 +899,17 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
  * Adaptive: if the per-byte key histograms show <= 5 active radix passes,
  * the plain key sort is cheaper; else 5 hash-byte passes + cleanup (with a
  * full-key-sort fallback on oversized dirty runs). */
+/* order_tag (optional out): 0 = result is FULL-KEY unsigned-ascending
+ * (narrow-key path, fallback, or trivial); 4 = result is (h32(key), key)
+ * unsigned-lexicographic (4 hash-byte passes + h32 cleanup). force_hbytes=4
+ * pins the hash order regardless of n (joins need a stable comparator);
+ * 0 = adaptive (reduce path). */
 hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
-                          uint64_t n, Ws &ws,
+                          uint64_t n, int force_hbytes, int *order_tag, Ws &ws,
                           const uint64_t **res_k, const uint64_t **res_v) {
     *res_k = in_k;
     *res_v = in_v;
+    if (order_tag) *order_tag = 0;
     if (n <= 1) return hipSuccess;
     uint32_t nb = nblocks_for(n);
 
@@ -904,9 +920,11 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
+    int *d_abort = (int *)ws.take(256);
     uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
-    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !h32buf)
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !d_abort || !h32buf)
         return hipErrorOutOfMemory;
+    HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
 
     static thread_local uint32_t hh[8 * 256];
     static thread_local uint32_t gb_host[8 * 256];
@@ -958,7 +976,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur, i == 0 ? in_v : nullptr,
                                      n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, true, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, true, in_pk, out_pk, df, "radix_scatter"));
             cur = dk;
         }
         *rk = cur;
@@ -976,15 +994,19 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(exact_hists(false, &active));
         if (active <= 5) { /* narrow keys: skipped key sort groups exactly */
             HIP_TRY(run_key_passes(res_k, res_v));
-            return hipSuccess;
+            int ab = 0;
+            HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
+            HIP_TRY(hipStreamSynchronize(s));
+            return ab ? hipErrorUnknown : hipSuccess;
         }
         /* sample lied; fall through to the hash path */
     }
 
     /* hash grouping: enough hash bytes to keep the expected bucket load
      * factor <= ~0.25 (4 bytes up to 2^30 rows, 5 above) — fewer passes,
-     * slightly busier cleanup */
-    const int hbytes = (n <= (1ULL << 30)) ? 4 : 5;
+     * slightly busier cleanup. force_hbytes pins the byte count (and with
+     * it the order contract) for joins. */
+    const int hbytes = force_hbytes ? force_hbytes : ((n <= (1ULL << 30)) ? 4 : 5);
     int active5 = 0;
     HIP_TRY(exact_hists(true, &active5));
     for (int i = 0; i < hbytes; ++i) {
@@ -996,7 +1018,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur_k, i == 0 ? in_v : nullptr,
                                  n, gbase_d + i * 256, desc, ticket,
                                  dk, dv, out_pk ? nullptr : h32buf,
-                                 true, in_pk, out_pk, df, "radix_scatter"));
+                                 d_abort, true, in_pk, out_pk, df, "radix_scatter"));
         cur_k = dk;
         cur_v = out_pk ? nullptr : dk + n;
     }
@@ -1016,6 +1038,11 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(exact_hists(false, &active));
         HIP_TRY(run_key_passes(&cur_k, &cur_v));
     }
+    int ab = 0;
+    HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (ab) return hipErrorUnknown; /* lookback bailed: fail loudly */
+    if (order_tag && !err) *order_tag = (hbytes == 4) ? 4 : 0;
     *res_k = cur_k;
     *res_v = cur_v;
     return hipSuccess;
@@ -1464,6 +1491,18 @@ hipError_t checksum_pairs(hipStream_t s, const int64_t *k, const int64_t *v,
     return hipStreamSynchronize(s);
 }
 
+hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
+                               uint64_t n, int *order_tag, Ws &ws) {
+    const uint64_t *rk, *rv;
+    HIP_TRY(group_sort_u64(s, (const uint64_t *)keys, (const uint64_t *)vals, n,
+                           /*force_hbytes=*/4, order_tag, ws, &rk, &rv));
+    if ((const uint64_t *)keys != rk) {
+        HIP_TRY(hipMemcpyAsync(keys, rk, n * 8, hipMemcpyDeviceToDevice, s));
+        HIP_TRY(hipMemcpyAsync(vals, rv, n * 8, hipMemcpyDeviceToDevice, s));
+    }
+    return hipSuccess;
+}
+
 size_t ws_bytes_for(uint64_t n) {
     uint64_t nb = nblocks_for(n ? n : 1);
     size_t b = 0;
@@ -1487,8 +1526,22 @@ size_t ws_bytes_for(uint64_t n) {
  * equal-key run in B (log2 nb probes; upper tree levels stay in L2/L3),
  * count + base, exclusive scan, then emit the cross product. */
 
+/* join comparator modes (must match how the sides were sorted):
+ *   0: signed key ascending (sort_by_key order; external callers)
+ *   1: unsigned key ascending (radix_sort_u64 signed_order=false)
+ *   2: (h32(key), key) unsigned-lexicographic (group order, tag 4) */
+__device__ __forceinline__ bool join_less(int mode, int64_t a, int64_t b) {
+    if (mode == 0) return a < b;
+    if (mode == 1) return (uint64_t)a < (uint64_t)b;
+    uint32_t ha = (uint32_t)vega_hash_u64((uint64_t)a);
+    uint32_t hb = (uint32_t)vega_hash_u64((uint64_t)b);
+    if (ha != hb) return ha < hb;
+    return (uint64_t)a < (uint64_t)b;
+}
+
 __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
-                             uint64_t nb, uint32_t *counts, uint32_t *b_lo) {
+                             uint64_t nb, int mode, uint32_t *counts,
+                             uint32_t *b_lo) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < na; i += stride) {
         int64_t k = ak[i];
@@ -1496,14 +1549,14 @@ __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
         uint64_t lo = 0, hi = nb;
         while (lo < hi) {
             uint64_t m = (lo + hi) >> 1;
-            if (bk[m] < k) lo = m + 1; else hi = m;
+            if (join_less(mode, bk[m], k)) lo = m + 1; else hi = m;
         }
         uint64_t lb = lo;
         /* upper bound */
         hi = nb;
         while (lo < hi) {
             uint64_t m = (lo + hi) >> 1;
-            if (bk[m] <= k) lo = m + 1; else hi = m;
+            if (!join_less(mode, k, bk[m])) lo = m + 1; else hi = m;
         }
         counts[i] = (uint32_t)(lo - lb);
         b_lo[i] = (uint32_t)lb;
@@ -1532,6 +1585,7 @@ __global__ void k_join_emit(const int64_t *ak, const int64_t *av, uint64_t na,
 
 hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint64_t na,
                        const int64_t *bk, const int64_t *bv, uint64_t nb,
+                       int mode,
                        int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                        uint64_t cap, uint64_t *h_nout, Ws &ws) {
     if (na == 0 || nb == 0) { *h_nout = 0; return hipSuccess; }
@@ -1542,7 +1596,8 @@ hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint
     uint32_t gb = nb_grid < 2048 ? nb_grid : 2048;
     {
         ProfScope ps("join_count", s);
-        hipLaunchKernelGGL(k_join_count, dim3(gb), dim3(BLOCK), 0, s, ak, na, bk, nb, counts, b_lo);
+        hipLaunchKernelGGL(k_join_count, dim3(gb), dim3(BLOCK), 0, s, ak, na, bk, nb,
+                           mode, counts, b_lo);
         HIP_TRY(hipGetLastError());
     }
     HIP_TRY(hipMemsetAsync(counts + na, 0, 4, s));

@@ -326,6 +326,27 @@ def dev_sort_pairs(keys_t, vals_t, ws_t):
            "dev_sort_pairs")
 
 
+def dev_group_pairs(keys_t, vals_t, ws_t):
+    """in-place grouping-order sort; returns the order tag (4 = (h32,key)
+    lex usable with dev_join_grouped mode 2; 0 = full unsigned-key order)"""
+    tag = ctypes.c_int(0)
+    _check(lib().vega_dev_group_pairs_i64(
+        _stream(), _t(keys_t), _t(vals_t), ctypes.c_uint64(keys_t.numel()),
+        ctypes.byref(tag), _t(ws_t), ctypes.c_size_t(ws_t.numel())), "dev_group_pairs")
+    return tag.value
+
+
+def dev_join_grouped(ak, av, bk, bv, order_mode, out_k, out_va, out_vb, ws_t):
+    nout = ctypes.c_uint64()
+    _check(lib().vega_dev_join_grouped(
+        _stream(), _t(ak), _t(av), ctypes.c_uint64(ak.numel()),
+        _t(bk), _t(bv), ctypes.c_uint64(bk.numel()), ctypes.c_int(order_mode),
+        _t(out_k), _t(out_va), _t(out_vb), ctypes.c_uint64(out_k.numel()),
+        ctypes.byref(nout), _t(ws_t), ctypes.c_size_t(ws_t.numel())),
+        "dev_join_grouped")
+    return nout.value
+
+
 def dev_join_sorted(ak, av, bk, bv, out_k, out_va, out_vb, ws_t):
     """sort-merge inner join of two KEY-SORTED sides; returns rows emitted"""
     nout = ctypes.c_uint64()
