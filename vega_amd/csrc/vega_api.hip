@@ -9,11 +9,14 @@
  * there is no CPU fallback anywhere in this library.
  */
 #include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
 #include <cstdio>
 #include <cstring>
 #include <map>
 #include <string>
+#include <vector>
 
+#include "../../include/vega_common.h"
 #include "../../include/vega_gpu.h"
 #include "vega_internal.h"
 
@@ -28,6 +31,11 @@ struct RddImpl {
     uint64_t alloc_rows = 0;
     uint32_t nparts = 1;
     bool sorted = false;
+    /* multi-GPU (ngpus > 1): per-device shards; d_k/d_v above are device 0's
+     * shard so the single-GPU code paths stay untouched for G == 1 */
+    std::vector<int64_t *> mk;
+    std::vector<void *> mv;
+    std::vector<uint64_t> mn;
 };
 
 struct vega_ctx {
@@ -38,6 +46,13 @@ struct vega_ctx {
     void *ws = nullptr;
     size_t ws_bytes = 0;
     char err[512] = {0};
+    /* multi-GPU (one process drives all G devices; vega local-mode shape):
+     * per-device streams, RCCL communicators, workspaces */
+    int ngpus = 1;
+    std::vector<hipStream_t> mstreams;
+    std::vector<ncclComm_t> comms;
+    std::vector<void *> mws;
+    std::vector<size_t> mws_bytes;
 };
 
 #define CTX_TRY(ctx, x)                                                        \
@@ -86,29 +101,79 @@ static int new_rdd(vega_ctx *c, uint64_t rows_alloc, int vtype, uint32_t nparts,
 extern "C" {
 
 int vega_gpu_init(int ngpus, vega_ctx_t **out) {
-    if (ngpus != 1) return VEGA_ERR_UNSUPPORTED; /* rank-per-GPU model */
     int ndev = 0;
     if (hipGetDeviceCount(&ndev) != hipSuccess || ndev < 1) return VEGA_ERR_HIP;
+    if (ngpus < 1 || ngpus > ndev || ngpus > 256) return VEGA_ERR_INVALID;
     vega_ctx *c = new vega_ctx();
+    c->ngpus = ngpus;
     (void)hipGetDevice(&c->device);
     if (hipStreamCreate(&c->stream) != hipSuccess) {
         delete c;
         return VEGA_ERR_HIP;
     }
+    if (ngpus > 1) {
+        /* one process drives all G devices over RCCL/xGMI (the in-process
+         * analogue of the reference's local scheduler) */
+        c->mstreams.resize(ngpus);
+        c->comms.resize(ngpus);
+        c->mws.assign(ngpus, nullptr);
+        c->mws_bytes.assign(ngpus, 0);
+        for (int g = 0; g < ngpus; ++g) {
+            if (hipSetDevice(g) != hipSuccess ||
+                hipStreamCreate(&c->mstreams[g]) != hipSuccess) {
+                delete c;
+                return VEGA_ERR_HIP;
+            }
+        }
+        if (ncclCommInitAll(c->comms.data(), ngpus, nullptr) != ncclSuccess) {
+            delete c;
+            return VEGA_ERR_HIP;
+        }
+        (void)hipSetDevice(0);
+    }
     *out = c;
     return VEGA_OK;
+}
+
+static void free_rdd_buffers(vega_ctx *c, RddImpl *r) {
+    if (c->ngpus > 1 && !r->mk.empty()) {
+        for (int g = 0; g < c->ngpus; ++g) {
+            (void)hipSetDevice(g);
+            if (r->mk[g]) (void)hipFree(r->mk[g]);
+            if (r->mv[g]) (void)hipFree(r->mv[g]);
+        }
+        (void)hipSetDevice(0);
+    } else {
+        if (r->d_k) (void)hipFree(r->d_k);
+        if (r->d_v) (void)hipFree(r->d_v);
+    }
+    if (r->d_v2) (void)hipFree(r->d_v2);
 }
 
 int vega_gpu_shutdown(vega_ctx_t *c) {
     if (!c) return VEGA_ERR_INVALID;
     (void)hipStreamSynchronize(c->stream);
+    for (int g = 0; g < (int)c->mstreams.size(); ++g) {
+        (void)hipSetDevice(g);
+        (void)hipStreamSynchronize(c->mstreams[g]);
+    }
     for (auto &kv : c->rdds) {
-        if (kv.second->d_k) (void)hipFree(kv.second->d_k);
-        if (kv.second->d_v) (void)hipFree(kv.second->d_v);
-        if (kv.second->d_v2) (void)hipFree(kv.second->d_v2);
+        free_rdd_buffers(c, kv.second);
         delete kv.second;
     }
     if (c->ws) (void)hipFree(c->ws);
+    for (int g = 0; g < (int)c->mws.size(); ++g) {
+        if (c->mws[g]) {
+            (void)hipSetDevice(g);
+            (void)hipFree(c->mws[g]);
+        }
+    }
+    for (auto &cm : c->comms) (void)ncclCommDestroy(cm);
+    for (int g = 0; g < (int)c->mstreams.size(); ++g) {
+        (void)hipSetDevice(g);
+        (void)hipStreamDestroy(c->mstreams[g]);
+    }
+    (void)hipSetDevice(c->device);
     (void)hipStreamDestroy(c->stream);
     delete c;
     return VEGA_OK;
@@ -122,9 +187,69 @@ int vega_gpu_synchronize(vega_ctx_t *c) {
 
 const char *vega_gpu_last_error(vega_ctx_t *c) { return c ? c->err : "null ctx"; }
 
+static int ensure_mws(vega_ctx *c, int g, uint64_t n) {
+    size_t need = ws_bytes_for(n);
+    if (c->mws_bytes[g] >= need) return VEGA_OK;
+    CTX_TRY(c, hipSetDevice(g));
+    if (c->mws[g]) (void)hipFree(c->mws[g]);
+    c->mws[g] = nullptr;
+    c->mws_bytes[g] = 0;
+    CTX_TRY(c, hipMalloc(&c->mws[g], need));
+    c->mws_bytes[g] = need;
+    return VEGA_OK;
+}
+
+/* new multi-sharded rdd: alloc_g rows per device */
+static int new_mrdd(vega_ctx *c, const std::vector<uint64_t> &alloc_g, int vtype,
+                    uint32_t nparts, RddImpl **out, vega_rdd_t *hout) {
+    RddImpl *r = new RddImpl();
+    r->vtype = vtype;
+    r->nparts = nparts ? nparts : 1;
+    r->mk.assign(c->ngpus, nullptr);
+    r->mv.assign(c->ngpus, nullptr);
+    r->mn.assign(c->ngpus, 0);
+    for (int g = 0; g < c->ngpus; ++g) {
+        uint64_t a = alloc_g[g] ? alloc_g[g] : 1;
+        CTX_TRY(c, hipSetDevice(g));
+        void *k = nullptr, *v = nullptr;
+        CTX_TRY(c, hipMalloc(&k, a * 8));
+        CTX_TRY(c, hipMalloc(&v, a * 8));
+        r->mk[g] = (int64_t *)k;
+        r->mv[g] = v;
+    }
+    (void)hipSetDevice(0);
+    uint64_t h = c->next_id++;
+    c->rdds[h] = r;
+    *out = r;
+    *hout = h;
+    return VEGA_OK;
+}
+
 static int make_rdd_common(vega_ctx *c, const int64_t *keys, const void *vals,
                            uint64_t n, uint32_t nparts, int vtype, vega_rdd_t *out) {
     if (!c || (!keys && n) || (!vals && n)) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) { /* shard rows [g*n/G,(g+1)*n/G) per device (a9) */
+        std::vector<uint64_t> sizes(c->ngpus);
+        for (int g = 0; g < c->ngpus; ++g)
+            sizes[g] = vega_slice_start(n, c->ngpus, g + 1) - vega_slice_start(n, c->ngpus, g);
+        RddImpl *r;
+        int rc = new_mrdd(c, sizes, vtype, nparts, &r, out);
+        if (rc) return rc;
+        for (int g = 0; g < c->ngpus; ++g) {
+            uint64_t lo = vega_slice_start(n, c->ngpus, g);
+            r->mn[g] = sizes[g];
+            if (!sizes[g]) continue;
+            CTX_TRY(c, hipSetDevice(g));
+            CTX_TRY(c, hipMemcpyAsync(r->mk[g], keys + lo, sizes[g] * 8,
+                                      hipMemcpyHostToDevice, c->mstreams[g]));
+            CTX_TRY(c, hipMemcpyAsync(r->mv[g], (const char *)vals + lo * 8, sizes[g] * 8,
+                                      hipMemcpyHostToDevice, c->mstreams[g]));
+        }
+        for (int g = 0; g < c->ngpus; ++g) CTX_TRY(c, hipStreamSynchronize(c->mstreams[g]));
+        (void)hipSetDevice(0);
+        r->n = n;
+        return VEGA_OK;
+    }
     RddImpl *r;
     int rc = new_rdd(c, n ? n : 1, vtype, nparts, &r, out);
     if (rc) return rc;
@@ -149,6 +274,24 @@ int vega_gpu_make_rdd_f64(vega_ctx_t *c, const int64_t *keys, const double *vals
 int vega_gpu_gen_rdd_uniform(vega_ctx_t *c, uint64_t n, uint64_t seed, int key_bits,
                              uint64_t start, uint32_t nparts, vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) {
+        std::vector<uint64_t> sizes(c->ngpus);
+        for (int g = 0; g < c->ngpus; ++g)
+            sizes[g] = vega_slice_start(n, c->ngpus, g + 1) - vega_slice_start(n, c->ngpus, g);
+        RddImpl *r;
+        int rc = new_mrdd(c, sizes, 0, nparts, &r, out);
+        if (rc) return rc;
+        for (int g = 0; g < c->ngpus; ++g) {
+            r->mn[g] = sizes[g];
+            if (!sizes[g]) continue;
+            CTX_TRY(c, hipSetDevice(g));
+            CTX_TRY(c, gen_uniform(c->mstreams[g], r->mk[g], (int64_t *)r->mv[g], sizes[g],
+                                   seed, key_bits, start + vega_slice_start(n, c->ngpus, g), false));
+        }
+        (void)hipSetDevice(0);
+        r->n = n;
+        return VEGA_OK;
+    }
     RddImpl *r;
     int rc = new_rdd(c, n ? n : 1, 0, nparts, &r, out);
     if (rc) return rc;
@@ -161,10 +304,109 @@ int vega_gpu_gen_rdd_uniform(vega_ctx_t *c, uint64_t n, uint64_t seed, int key_b
  * partition count (pair_rdd.rs:54-80); the collected RESULT is invariant to
  * it (each key lands in exactly one reduce partition either way), so the
  * single-GPU engine computes the global aggregate directly. */
+/* G>1 reduce: per-device radix partition into G buckets (bucket g2 owned by
+ * device g2, replacing the MapOutputTracker+HTTP pull with host-side counts
+ * + one RCCL grouped send/recv all-to-all-v over xGMI), then per-device
+ * grouping sort + segmented aggregate. */
+static int reduce_multi(vega_ctx *c, RddImpl *r, int op, uint32_t nparts,
+                        vega_rdd_t *out) {
+    const int G = c->ngpus;
+    if ((op == VEGA_OP_SUM_F64) && r->vtype != 1) return VEGA_ERR_INVALID;
+    /* per-device partition into G buckets */
+    std::vector<int64_t *> pk(G, nullptr), pv(G, nullptr);
+    std::vector<std::vector<uint64_t>> counts(G, std::vector<uint64_t>(G, 0));
+    for (int g = 0; g < G; ++g) {
+        int rc = ensure_mws(c, g, r->mn[g] + 1024);
+        if (rc) return rc;
+        CTX_TRY(c, hipSetDevice(g));
+        uint64_t a = r->mn[g] ? r->mn[g] : 1;
+        void *k = nullptr, *v = nullptr;
+        CTX_TRY(c, hipMalloc(&k, a * 8));
+        CTX_TRY(c, hipMalloc(&v, a * 8));
+        pk[g] = (int64_t *)k;
+        pv[g] = (int64_t *)v;
+        Ws ws(c->mws[g], c->mws_bytes[g]);
+        CTX_TRY(c, hash_partition(c->mstreams[g], (const uint64_t *)r->mk[g],
+                                  (const uint64_t *)r->mv[g], r->mn[g], (uint32_t)G,
+                                  (uint64_t *)pk[g], (uint64_t *)pv[g],
+                                  counts[g].data(), ws));
+    }
+    /* recv shard sizes per destination device */
+    std::vector<uint64_t> rn(G, 0);
+    for (int g = 0; g < G; ++g)
+        for (int p = 0; p < G; ++p) rn[p] += counts[g][p];
+    std::vector<int64_t *> rk(G, nullptr), rv(G, nullptr);
+    for (int p = 0; p < G; ++p) {
+        CTX_TRY(c, hipSetDevice(p));
+        uint64_t a = rn[p] ? rn[p] : 1;
+        void *k = nullptr, *v = nullptr;
+        CTX_TRY(c, hipMalloc(&k, a * 8));
+        CTX_TRY(c, hipMalloc(&v, a * 8));
+        rk[p] = (int64_t *)k;
+        rv[p] = (int64_t *)v;
+    }
+    /* grouped all-to-all-v (keys, then values) */
+    for (int pass = 0; pass < 2; ++pass) {
+        if (ncclGroupStart() != ncclSuccess) return VEGA_ERR_HIP;
+        std::vector<uint64_t> roff(G, 0);
+        for (int g = 0; g < G; ++g) {
+            uint64_t soff = 0;
+            for (int p = 0; p < G; ++p) {
+                uint64_t cnt = counts[g][p];
+                const int64_t *sbase = pass == 0 ? pk[g] : pv[g];
+                int64_t *rbase = pass == 0 ? rk[p] : rv[p];
+                if (cnt) {
+                    if (ncclSend(sbase + soff, cnt, ncclInt64, p, c->comms[g],
+                                 c->mstreams[g]) != ncclSuccess ||
+                        ncclRecv(rbase + roff[p], cnt, ncclInt64, g, c->comms[p],
+                                 c->mstreams[p]) != ncclSuccess)
+                        return VEGA_ERR_HIP;
+                }
+                soff += cnt;
+                roff[p] += cnt;
+            }
+        }
+        if (ncclGroupEnd() != ncclSuccess) return VEGA_ERR_HIP;
+    }
+    /* per-device grouping sort + segmented aggregate */
+    std::vector<uint64_t> alloc_out(G);
+    for (int p = 0; p < G; ++p) alloc_out[p] = rn[p] ? rn[p] : 1;
+    RddImpl *o;
+    int rc = new_mrdd(c, alloc_out, op == VEGA_OP_SUM_F64 ? 1 : 0, nparts, &o, out);
+    if (rc) return rc;
+    uint64_t total = 0;
+    for (int p = 0; p < G; ++p) {
+        rc = ensure_mws(c, p, rn[p] + 1024);
+        if (rc) return rc;
+        CTX_TRY(c, hipSetDevice(p));
+        Ws ws(c->mws[p], c->mws_bytes[p]);
+        const uint64_t *sk, *sv;
+        CTX_TRY(c, group_sort_u64(c->mstreams[p], (const uint64_t *)rk[p],
+                                  (const uint64_t *)rv[p], rn[p], ws, &sk, &sv));
+        uint64_t nout = 0;
+        CTX_TRY(c, seg_reduce(c->mstreams[p], sk, sv, rn[p], op,
+                              (uint64_t *)o->mk[p], o->mv[p], &nout, ws));
+        o->mn[p] = nout;
+        total += nout;
+    }
+    for (int g = 0; g < G; ++g) { /* transients */
+        CTX_TRY(c, hipSetDevice(g));
+        (void)hipFree(pk[g]);
+        (void)hipFree(pv[g]);
+        (void)hipFree(rk[g]);
+        (void)hipFree(rv[g]);
+    }
+    (void)hipSetDevice(0);
+    o->n = total;
+    o->sorted = false;
+    return VEGA_OK;
+}
+
 static int reduce_common(vega_ctx *c, vega_rdd_t rdd, int op, uint32_t nparts,
                          vega_rdd_t *out) {
     RddImpl *r = get_rdd(c, rdd);
     if (!r) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return reduce_multi(c, r, op, nparts, out);
     /* SUM_F64 needs f64 values; COUNT ignores values; the i64 ops need i64 */
     if (op == VEGA_OP_SUM_F64 && r->vtype != 1) return VEGA_ERR_INVALID;
     if (op != VEGA_OP_SUM_F64 && op != VEGA_OP_COUNT && r->vtype == 1) return VEGA_ERR_INVALID;
@@ -206,6 +448,7 @@ int vega_gpu_distinct(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts, vega_rdd_t
 int vega_gpu_count_by_value(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts,
                             vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED; /* G>1: north-star ops only (this branch) */
     RddImpl *r = get_rdd(c, rdd);
     if (!r || r->vtype != 0) return VEGA_ERR_INVALID;
     int rc = ensure_ws(c, r->n);
@@ -226,6 +469,7 @@ int vega_gpu_count_by_value(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts,
 
 int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED; /* G>1: north-star ops only (this branch) */
     RddImpl *r = get_rdd(c, rdd);
     if (!r) return VEGA_ERR_INVALID;
     int rc = ensure_ws(c, r->n);
@@ -250,6 +494,7 @@ int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
 int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
                   vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED; /* G>1: north-star ops only (this branch) */
     RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
     if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
     uint64_t nmax = ra->n > rb->n ? ra->n : rb->n;
@@ -358,6 +603,7 @@ int vega_gpu_collect_join(vega_ctx_t *c, vega_rdd_t rdd, int64_t *keys,
 int vega_gpu_map(vega_ctx_t *c, vega_rdd_t rdd, vega_map_op_t op, int64_t p0,
                  vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED; /* G>1: north-star ops only (this branch) */
     RddImpl *r = get_rdd(c, rdd);
     if (!r || r->vtype != 0) return VEGA_ERR_INVALID;
     RddImpl *o;
@@ -372,6 +618,7 @@ int vega_gpu_map(vega_ctx_t *c, vega_rdd_t rdd, vega_map_op_t op, int64_t p0,
 int vega_gpu_filter(vega_ctx_t *c, vega_rdd_t rdd, vega_pred_t pred,
                     int64_t p0, int64_t p1, vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED; /* G>1: north-star ops only (this branch) */
     RddImpl *r = get_rdd(c, rdd);
     if (!r || r->vtype != 0) return VEGA_ERR_INVALID;
     int rc = ensure_ws(c, r->n);
@@ -402,6 +649,23 @@ int vega_gpu_collect(vega_ctx_t *c, vega_rdd_t rdd, int64_t *keys, void *vals, u
         return VEGA_OK;
     }
     if (*n < r->n) return VEGA_ERR_CAP;
+    if (c->ngpus > 1 && !r->mk.empty()) { /* concat device shards */
+        uint64_t off = 0;
+        for (int g = 0; g < c->ngpus; ++g) {
+            if (!r->mn[g]) continue;
+            CTX_TRY(c, hipSetDevice(g));
+            CTX_TRY(c, hipMemcpyAsync(keys + off, r->mk[g], r->mn[g] * 8,
+                                      hipMemcpyDeviceToHost, c->mstreams[g]));
+            if (vals)
+                CTX_TRY(c, hipMemcpyAsync((char *)vals + off * 8, r->mv[g], r->mn[g] * 8,
+                                          hipMemcpyDeviceToHost, c->mstreams[g]));
+            off += r->mn[g];
+        }
+        for (int g = 0; g < c->ngpus; ++g) CTX_TRY(c, hipStreamSynchronize(c->mstreams[g]));
+        (void)hipSetDevice(0);
+        *n = off;
+        return VEGA_OK;
+    }
     *n = r->n;
     if (r->n) {
         CTX_TRY(c, hipMemcpyAsync(keys, r->d_k, r->n * 8, hipMemcpyDeviceToHost, c->stream));
@@ -416,9 +680,7 @@ int vega_gpu_free_rdd(vega_ctx_t *c, vega_rdd_t rdd) {
     auto it = c->rdds.find(rdd);
     if (it == c->rdds.end()) return VEGA_ERR_INVALID;
     CTX_TRY(c, hipStreamSynchronize(c->stream));
-    if (it->second->d_k) (void)hipFree(it->second->d_k);
-    if (it->second->d_v) (void)hipFree(it->second->d_v);
-    if (it->second->d_v2) (void)hipFree(it->second->d_v2);
+    free_rdd_buffers(c, it->second);
     delete it->second;
     c->rdds.erase(it);
     return VEGA_OK;

@@ -82,9 +82,9 @@ def _check(rc, what, ctx=None):
 class VegaContext:
     """Mirrors Context (context.rs:147-164) + PairRdd ops for one GPU."""
 
-    def __init__(self):
+    def __init__(self, ngpus=1):
         self._c = ctypes.c_void_p()
-        _check(lib().vega_gpu_init(1, ctypes.byref(self._c)), "init")
+        _check(lib().vega_gpu_init(ngpus, ctypes.byref(self._c)), "init")
 
     def close(self):
         if self._c:
