@@ -470,6 +470,17 @@ int vega_dev_join_sorted(void *stream, const int64_t *ak, const int64_t *av, uin
     return e == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
 }
 
+/* diagnostic (VEGA_PHASE_PROF=1 builds/runs): per-phase shader-cycle sums of
+ * the onesweep scatter — 0 prefetch, 1 rank, 2 publish+starts, 3 lookback,
+ * 4 reorder, 5 writeout. */
+int vega_phase_prof_read(unsigned long long out[8], int reset) {
+    unsigned long long *b = vega::phase_prof_buf();
+    if (!b) return VEGA_ERR_UNSUPPORTED;
+    if (hipMemcpy(out, b, 8 * 8, hipMemcpyDeviceToHost) != hipSuccess) return VEGA_ERR_HIP;
+    if (reset) (void)hipMemset(b, 0, 8 * 8);
+    return VEGA_OK;
+}
+
 int vega_dev_checksum_pairs(void *stream, const int64_t *keys, const int64_t *vals,
                             uint64_t n, uint64_t *h_sum, void *d_ws, size_t ws_bytes) {
     Ws ws(d_ws, ws_bytes);

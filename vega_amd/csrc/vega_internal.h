@@ -96,6 +96,9 @@ hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint
 
 size_t ws_bytes_for(uint64_t n);
 
+/* diagnostic phase-cycle buffer (VEGA_PHASE_PROF=1), else nullptr */
+unsigned long long *phase_prof_buf();
+
 } // namespace vega
 
 #endif
