@@ -382,7 +382,7 @@ static int reduce_multi(vega_ctx *c, RddImpl *r, int op, uint32_t nparts,
         Ws ws(c->mws[p], c->mws_bytes[p]);
         const uint64_t *sk, *sv;
         CTX_TRY(c, group_sort_u64(c->mstreams[p], (const uint64_t *)rk[p],
-                                  (const uint64_t *)rv[p], rn[p], ws, &sk, &sv));
+                                  (const uint64_t *)rv[p], rn[p], 0, nullptr, ws, &sk, &sv));
         uint64_t nout = 0;
         CTX_TRY(c, seg_reduce(c->mstreams[p], sk, sv, rn[p], op,
                               (uint64_t *)o->mk[p], o->mv[p], &nout, ws));
@@ -798,6 +798,17 @@ int vega_dev_join_grouped(void *stream, const int64_t *ak, const int64_t *av, ui
     hipError_t e = join_sorted((hipStream_t)stream, ak, av, na, bk, bv, nb, order_mode,
                                out_k, out_va, out_vb, cap, h_nout, ws);
     return e == hipSuccess ? VEGA_OK : VEGA_ERR_HIP;
+}
+
+/* diagnostic (VEGA_PHASE_PROF=1 builds/runs): per-phase shader-cycle sums of
+ * the onesweep scatter — 0 prefetch, 1 rank, 2 publish+starts, 3 lookback,
+ * 4 reorder, 5 writeout. */
+int vega_phase_prof_read(unsigned long long out[8], int reset) {
+    unsigned long long *b = vega::phase_prof_buf();
+    if (!b) return VEGA_ERR_UNSUPPORTED;
+    if (hipMemcpy(out, b, 8 * 8, hipMemcpyDeviceToHost) != hipSuccess) return VEGA_ERR_HIP;
+    if (reset) (void)hipMemset(b, 0, 8 * 8);
+    return VEGA_OK;
 }
 
 int vega_dev_checksum_pairs(void *stream, const int64_t *keys, const int64_t *vals,

@@ -105,6 +105,9 @@ hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
 
 size_t ws_bytes_for(uint64_t n);
 
+/* diagnostic phase-cycle buffer (VEGA_PHASE_PROF=1), else nullptr */
+unsigned long long *phase_prof_buf();
+
 } // namespace vega
 
 #endif

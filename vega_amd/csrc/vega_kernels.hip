@@ -471,7 +471,12 @@ template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
 __global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
-    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, int *d_abort, DF df) {
+    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, int *d_abort,
+    unsigned long long *phc, DF df) {
+    /* phc (diagnostic builds, VEGA_PHASE_PROF=1): per-phase shader-cycle
+     * sums, one sample per wave — phases: 0 prefetch, 1 rank, 2 publish+
+     * starts, 3 lookback, 4 reorder, 5 writeout (s_memtime; the microarch
+     * guide prices the instrumentation itself at ~+11% wave cycles). */
     /* 512 threads = 8 waves per block (16 waves/CU at 2 blocks): each wave
      * ranks a 512-row chunk of the 4096-row tile. IN_PK/OUT_PK: interleaved
      * (k,v) rows — one 16-B vector access per row. */
@@ -487,6 +492,13 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t lower = ((uint64_t)1 << lane) - 1;
+    unsigned long long tprev = phc ? __builtin_amdgcn_s_memtime() : 0;
+#define VEGA_PHASE_MARK(i)                                                     \
+    if (phc) {                                                                 \
+        unsigned long long tnow = __builtin_amdgcn_s_memtime();                \
+        if (lane == 0) atomicAdd(&phc[i], tnow - tprev);                       \
+        tprev = tnow;                                                          \
+    }
 
     if (t == 0) vbp[0] = atomicAdd(ticket, 1u); /* scheduling-ordered tile id */
     for (int i = t; i < SW * 256; i += SB) whist[i] = 0;
@@ -534,6 +546,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
             }
         }
     }
+    VEGA_PHASE_MARK(0)
     /* ranking: register-only rounds over the prefetched chunk */
 #pragma unroll
     for (int r = 0; r < SIPT; ++r) {
@@ -550,6 +563,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         rank[r] = base + (uint32_t)__popcll(m & lower);
     }
     __syncthreads();
+    VEGA_PHASE_MARK(1)
 
     /* digit threads (t<256): counts -> publish AGG, local starts, per-wave
      * offsets, lookback */
@@ -585,6 +599,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
             whist[wv * 256 + t] = run;
             run += cw[wv];
         }
+        VEGA_PHASE_MARK(2)
         unsigned long long excl_tiles = 0;
         if (vb > 0) {
             gdesc_t *col = (gdesc_t *)(desc + t);
@@ -620,6 +635,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
     }
     __syncthreads();
+    VEGA_PHASE_MARK(3)
 
     /* reorder into LDS at the stable tile-local position */
     {
@@ -637,6 +653,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         }
     }
     __syncthreads();
+    VEGA_PHASE_MARK(4)
 
     /* write out: digit-contiguous global writes */
     for (uint32_t p = t; p < tile_n; p += SB) {
@@ -660,6 +677,24 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 h32_out[gpos] = (uint32_t)vega_hash_u64(k);
         }
     }
+    VEGA_PHASE_MARK(5)
+#undef VEGA_PHASE_MARK
+}
+
+/* diagnostic per-phase cycle sums (VEGA_PHASE_PROF=1): lazily allocated,
+ * read/reset via vega_phase_prof_read() */
+static unsigned long long *g_phase_buf = nullptr;
+unsigned long long *phase_prof_buf() {
+    static bool checked = false;
+    if (!checked) {
+        checked = true;
+        const char *e = getenv("VEGA_PHASE_PROF");
+        if (e && e[0] == '1') {
+            (void)hipMalloc(&g_phase_buf, 8 * 8);
+            (void)hipMemset(g_phase_buf, 0, 8 * 8);
+        }
+    }
+    return g_phase_buf;
 }
 
 template <class DF>
@@ -669,6 +704,7 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out,
                                    int *d_abort, bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
+    unsigned long long *phc = phase_prof_buf();
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
@@ -676,19 +712,19 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
         hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, phc, df);
     } else if (!in_pk && !out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, phc, df);
     } else if (!in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, phc, df);
     } else if (in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, phc, df);
     } else {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, phc, df);
     }
     return hipGetLastError();
 }
