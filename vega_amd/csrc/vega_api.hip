@@ -61,7 +61,8 @@ struct vega_ctx {
         if (_e != hipSuccess) {                                                \
             snprintf((ctx)->err, sizeof (ctx)->err, "%s:%d: %s", __FILE__,     \
                      __LINE__, hipGetErrorString(_e));                         \
-            return VEGA_ERR_HIP;                                               \
+            return _e == hipErrorNotSupported ? VEGA_ERR_UNSUPPORTED           \
+                                              : VEGA_ERR_HIP;                  \
         }                                                                      \
     } while (0)
 
@@ -437,9 +438,26 @@ int vega_gpu_group_count(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts, vega_rd
     return reduce_common(c, rdd, VEGA_OP_COUNT, nparts, out);
 }
 
+/* distinct (rdd.rs:501-531): the deduplicated key set. The reference's
+ * distinct returns ELEMENTS (map x->(x,None) + reduce_by_key + unwrap); here
+ * the element column is the key, so values of the result are zeroed — the
+ * result is the key set, not (key, count) (use group_count for counts). */
 int vega_gpu_distinct(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts, vega_rdd_t *out) {
     if (!c) return VEGA_ERR_INVALID;
-    return reduce_common(c, rdd, VEGA_OP_COUNT, nparts, out);
+    int rc = reduce_common(c, rdd, VEGA_OP_COUNT, nparts, out);
+    if (rc) return rc;
+    RddImpl *o = get_rdd(c, *out);
+    if (o && c->ngpus > 1 && !o->mk.empty()) {
+        for (int g = 0; g < c->ngpus; ++g) {
+            if (!o->mn[g]) continue;
+            CTX_TRY(c, hipSetDevice(g));
+            CTX_TRY(c, hipMemsetAsync(o->mv[g], 0, o->mn[g] * 8, c->mstreams[g]));
+        }
+        (void)hipSetDevice(0);
+    } else if (o && o->n) {
+        CTX_TRY(c, hipMemsetAsync(o->d_v, 0, o->n * 8, c->stream));
+    }
+    return VEGA_OK;
 }
 
 /* count_by_value (rdd.rs:449-459): group-count with the VALUE column as the

@@ -797,6 +797,7 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     *res_k = in_k;
     *res_v = in_v;
     if (n <= 1) return hipSuccess;
+    if (n >= (1ULL << 32)) return hipErrorNotSupported; /* u32 hist/scan limit (vega_gpu.h) */
     uint32_t nb = nblocks_for(n);
 
     uint64_t *pA = (uint64_t *)ws.take(n * 16);
@@ -887,8 +888,26 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
  * contract needs). h32 is never rewritten, so cross-thread reads stay
  * consistent while a run's owner permutes k/v (keys within a run keep the
  * run's h32 by definition). */
+/* Two cleanup CONTRACTS:
+ *   strict=1 (joins, order_tag consumers): every equal-h32 run must end
+ *     fully key-sorted — the (h32, key) lexicographic order the join
+ *     comparator binary-searches.
+ *   strict=0 (reduce path): only "equal keys adjacent" is needed. A run of
+ *     length 2 needs NOTHING (two equal keys are already adjacent; two
+ *     distinct keys have nothing to group), and a longer run is fine as long
+ *     as it has <= 2 maximal equal-key segments ([A..A,B..B] groups
+ *     correctly; [A,B,A] does not). With mostly-distinct keys nearly every
+ *     hash-colliding pair is a distinct-key pair, so this skips ~95% of the
+ *     insertion-sort work the strict contract would do.
+ * Serial walks are BOUNDED: a run extending past
+ * CLEANUP_WALK_CAP rows is appended to a worklist and handled by
+ * k_group_cleanup_long with a whole workgroup (parallel end-scan + parallel
+ * break count), so a Zipf hot key can never put an O(run) walk on one lane. */
+#define CLEANUP_WALK_CAP 1024
+#define CLEANUP_WL_CAP 65536
 __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
-                                uint64_t n, int *err) {
+                                uint64_t n, int strict, int *err,
+                                unsigned long long *wl, uint32_t *wl_count) {
     uint64_t nchunks = (n + 3) / 4;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
@@ -908,14 +927,36 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
             /* peek in-window: single-row runs need no global traffic */
             if (j + 1 < m && hh[j + 2] != hh[j + 1]) continue;
             uint64_t je = gi + 1;
-            while (je < n && h32[je] == hh[j + 1]) je++;
-            if (je - gi == 1) continue;
-            /* run of length >= 2: check keys */
-            uint64_t k0 = k[gi];
-            bool dirty = false;
-            for (uint64_t x = gi + 1; x < je && !dirty; x++) dirty = (k[x] != k0);
-            if (!dirty) continue;
-            if (je - gi > 64) { *err = 1; continue; }
+            uint64_t wcap = gi + CLEANUP_WALK_CAP;
+            while (je < n && je < wcap && h32[je] == hh[j + 1]) je++;
+            if (je == wcap && je < n && h32[je] == hh[j + 1]) {
+                /* long run: hand to the cooperative kernel */
+                uint32_t slot = atomicAdd(wl_count, 1u);
+                if (slot < CLEANUP_WL_CAP) wl[slot] = gi;
+                else *err = 1; /* worklist full: full-key-sort fallback */
+                continue;
+            }
+            uint64_t len = je - gi;
+            if (len == 1) continue;
+            if (!strict && len == 2) continue; /* nothing to group either way */
+            bool fix;
+            if (strict) {
+                uint64_t k0 = k[gi];
+                bool dirty = false;
+                for (uint64_t x = gi + 1; x < je && !dirty; x++) dirty = (k[x] != k0);
+                fix = dirty;
+            } else {
+                uint32_t breaks = 0;
+                uint64_t prev = k[gi];
+                for (uint64_t x = gi + 1; x < je && breaks < 2; x++) {
+                    uint64_t cx = k[x];
+                    breaks += cx != prev;
+                    prev = cx;
+                }
+                fix = breaks > 1; /* >2 segments: some key may repeat non-adjacently */
+            }
+            if (!fix) continue;
+            if (len > 64) { *err = 1; continue; }
             for (uint64_t x = gi + 1; x < je; x++) {
                 uint64_t kx = k[x], vx = v[x];
                 uint64_t y = x;
@@ -928,6 +969,48 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
                 v[y] = vx;
             }
         }
+    }
+}
+
+/* cooperative handler for runs longer than CLEANUP_WALK_CAP: one workgroup
+ * per worklist entry — parallel strided end-scan, then a parallel break
+ * count. A long run that actually needs fixing (>2 segments relaxed; any
+ * mismatch strict) is too big for an in-place insertion sort, so it sets
+ * *err and the caller falls back to the full key sort. The common long run
+ * (a hot key, all rows equal) passes with zero breaks. */
+__global__ void k_group_cleanup_long(const uint64_t *k, const uint32_t *h32,
+                                     uint64_t n, int strict, int *err,
+                                     const unsigned long long *wl,
+                                     const uint32_t *wl_count) {
+    __shared__ unsigned long long s_end;
+    __shared__ unsigned int s_breaks;
+    uint32_t cnt = *wl_count;
+    if (cnt > CLEANUP_WL_CAP) cnt = CLEANUP_WL_CAP;
+    for (uint32_t e = blockIdx.x; e < cnt; e += gridDim.x) {
+        uint64_t gi = wl[e];
+        uint32_t h = h32[gi];
+        if (threadIdx.x == 0) { s_end = n; s_breaks = 0; }
+        __syncthreads();
+        for (uint64_t base = gi + 1; base < s_end; base += blockDim.x) {
+            uint64_t x = base + threadIdx.x;
+            if (x < n && h32[x] != h) atomicMin(&s_end, (unsigned long long)x);
+            __syncthreads();
+        }
+        uint64_t je = s_end;
+        uint64_t k0 = k[gi];
+        for (uint64_t base = gi + 1; base < je && s_breaks < 2; base += blockDim.x) {
+            uint64_t x = base + threadIdx.x;
+            if (x < je) {
+                bool br = strict ? (k[x] != k0) : (k[x] != k[x - 1]);
+                if (br) atomicAdd(&s_breaks, 1u);
+            }
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) {
+            bool bad = strict ? (s_breaks != 0) : (s_breaks > 1);
+            if (bad) *err = 1;
+        }
+        __syncthreads();
     }
 }
 
@@ -947,6 +1030,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     *res_v = in_v;
     if (order_tag) *order_tag = 0;
     if (n <= 1) return hipSuccess;
+    if (n >= (1ULL << 32)) return hipErrorNotSupported; /* u32 hist/scan limit (vega_gpu.h) */
     uint32_t nb = nblocks_for(n);
 
     uint64_t *pA = (uint64_t *)ws.take(n * 16);
@@ -958,9 +1042,16 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     int *d_err = (int *)ws.take(256);
     int *d_abort = (int *)ws.take(256);
     uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
-    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !d_abort || !h32buf)
+    unsigned long long *wl = (unsigned long long *)ws.take(CLEANUP_WL_CAP * 8);
+    uint32_t *wl_count = (uint32_t *)ws.take(256);
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !d_abort ||
+        !h32buf || !wl || !wl_count)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
+    /* strict order (full (h32,key) lex within runs) only when a caller will
+     * binary-search the result (joins/cogroup: force_hbytes or order_tag);
+     * the reduce path needs only equal-keys-adjacent (relaxed) */
+    const int strict = (force_hbytes != 0 || order_tag != nullptr) ? 1 : 0;
 
     static thread_local uint32_t hh[8 * 256];
     static thread_local uint32_t gb_host[8 * 256];
@@ -1059,11 +1150,16 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         cur_v = out_pk ? nullptr : dk + n;
     }
     HIP_TRY(hipMemsetAsync(d_err, 0, 4, s));
+    HIP_TRY(hipMemsetAsync(wl_count, 0, 4, s));
     {
         ProfScope ps("group_cleanup", s);
         uint32_t gb = nb < 2048 ? nb : 2048;
         hipLaunchKernelGGL(k_group_cleanup, dim3(gb), dim3(BLOCK), 0, s,
-                           (uint64_t *)cur_k, (uint64_t *)cur_v, h32buf, n, d_err);
+                           (uint64_t *)cur_k, (uint64_t *)cur_v, h32buf, n, strict,
+                           d_err, wl, wl_count);
+        HIP_TRY(hipGetLastError());
+        hipLaunchKernelGGL(k_group_cleanup_long, dim3(512), dim3(BLOCK), 0, s,
+                           cur_k, h32buf, n, strict, d_err, wl, wl_count);
         HIP_TRY(hipGetLastError());
     }
     int err = 0;
@@ -1112,11 +1208,19 @@ __global__ void k_head_count(const uint64_t *k, uint64_t n, uint32_t *hc) {
  * unrolled sequential loads vectorize and every 64 B line is fully consumed
  * by exactly one lane, so the pattern is bandwidth-clean and occupancy is
  * not LDS-bound. */
+/* OP==2 (SUM_F64) is DETERMINISTIC: no atomics anywhere. Interior runs are
+ * stored exclusively by the chunk that starts them; each chunk's leading
+ * partial (the part of the run open at its start) goes to
+ * (lead_seg, lead_part)[chunk], and k_f64_seg_combine folds those in chunk
+ * order — a fixed summation shape for a given n, so results are bit-stable
+ * run to run (and within 1e-6 rel of the reference's sequential merge,
+ * pair_rdd.rs:74-78). */
 template <int OP>
 __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     const uint64_t *__restrict__ k, const void *__restrict__ vv, uint64_t n,
     const uint32_t *__restrict__ head_base, int64_t *__restrict__ out_k,
-    void *__restrict__ out_vv) {
+    void *__restrict__ out_vv, uint32_t *__restrict__ lead_seg,
+    double *__restrict__ lead_part) {
     __shared__ uint32_t wsc[BLOCK / 64];
     constexpr bool NEED_V = (OP != 1);
 
@@ -1170,6 +1274,48 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     __syncthreads();
     uint32_t excl = inc - cnt;
     for (int i = 0; i < w; ++i) excl += wsc[i];
+
+    if (OP == 2) { /* deterministic f64 path (see kernel comment) */
+        uint64_t chunk_id = (uint64_t)blockIdx.x * BLOCK + t;
+        int64_t sid = (int64_t)head_base[blockIdx.x] + excl - 1; /* run open at chunk start */
+        if (c0g >= n) {
+            lead_seg[chunk_id] = 0xFFFFFFFFu;
+            return;
+        }
+        int nvalid = (int)((n - c0g < IPT) ? (n - c0g) : IPT);
+        int fh = -1;
+        for (int j = 0; j < nvalid; ++j) {
+            uint64_t pk2 = (j > 0) ? kk[j - 1] : prev;
+            if (c0g + j == 0 || kk[j] != pk2) { fh = j; break; }
+        }
+        int le = (fh >= 0) ? fh : nvalid; /* rows belonging to the open run */
+        if (le > 0) {
+            double lead = 0.0;
+            for (int j = 0; j < le; ++j) lead += __longlong_as_double((long long)sv[j]);
+            lead_seg[chunk_id] = (uint32_t)sid;
+            lead_part[chunk_id] = lead;
+        } else {
+            lead_seg[chunk_id] = 0xFFFFFFFFu;
+        }
+        if (fh >= 0) {
+            double acc = 0.0;
+            for (int j = fh; j < nvalid; ++j) {
+                uint64_t pk2 = (j > 0) ? kk[j - 1] : prev;
+                bool head = (c0g + j == 0) || (kk[j] != pk2);
+                if (head) {
+                    if (j > fh) ((double *)out_vv)[sid] = acc; /* interior run: exclusive */
+                    sid++;
+                    out_k[sid] = (int64_t)kk[j];
+                    acc = 0.0;
+                }
+                acc += __longlong_as_double((long long)sv[j]);
+            }
+            /* last run started in this chunk: its base value (later chunks'
+             * contributions are folded in by k_f64_seg_combine) */
+            ((double *)out_vv)[sid] = acc;
+        }
+        return;
+    }
 
     /* Fast path: a fully-distinct interior chunk whose last run does not
      * continue into the next chunk (the dominant C1 shape) emits its 16
@@ -1268,12 +1414,39 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     }
 }
 
+/* fold each chunk's leading f64 partial into its run's accumulator, walking
+ * the (contiguous) chunk range of each run in chunk order — fixed summation
+ * shape, deterministic. One thread per range start; ranges are disjoint so
+ * the += is exclusive. */
+__global__ void k_f64_seg_combine(const uint32_t *lead_seg, const double *lead_part,
+                                  uint64_t nchunks, double *out_v) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
+        uint32_t sg = lead_seg[c];
+        if (sg == 0xFFFFFFFFu) continue;
+        if (c > 0 && lead_seg[c - 1] == sg) continue; /* not the range start */
+        double acc = 0.0;
+        uint64_t x = c;
+        while (x < nchunks && lead_seg[x] == sg) { acc += lead_part[x]; ++x; }
+        out_v[sg] += acc;
+    }
+}
+
 hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t n,
                       int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws) {
     if (n == 0) { *h_nout = 0; return hipSuccess; }
+    if (n >= (1ULL << 32)) return hipErrorNotSupported; /* u32 head scan limit */
     uint32_t nb = nblocks_for(n);
     uint32_t *hc = (uint32_t *)ws.take(((size_t)nb + 1) * 4);
     if (!hc) return hipErrorOutOfMemory;
+    uint64_t nchunks = (uint64_t)nb * BLOCK;
+    uint32_t *lead_seg = nullptr;
+    double *lead_part = nullptr;
+    if (op == 2) {
+        lead_seg = (uint32_t *)ws.take(nchunks * 4);
+        lead_part = (double *)ws.take(nchunks * 8);
+        if (!lead_seg || !lead_part) return hipErrorOutOfMemory;
+    }
     {
         ProfScope ps("head_count", s);
         hipLaunchKernelGGL(k_head_count, dim3(nb), dim3(BLOCK), 0, s, k, n, hc);
@@ -1300,13 +1473,20 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
         ProfScope ps("seg_emit", s);
         size_t sh = 0;
         switch (op) {
-        case 0: hipLaunchKernelGGL(k_seg_emit<0>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v); break;
-        case 1: hipLaunchKernelGGL(k_seg_emit<1>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v); break;
-        case 2: hipLaunchKernelGGL(k_seg_emit<2>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v); break;
-        case 3: hipLaunchKernelGGL(k_seg_emit<3>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v); break;
-        case 4: hipLaunchKernelGGL(k_seg_emit<4>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v); break;
+        case 0: hipLaunchKernelGGL(k_seg_emit<0>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
+        case 1: hipLaunchKernelGGL(k_seg_emit<1>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
+        case 2: hipLaunchKernelGGL(k_seg_emit<2>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
+        case 3: hipLaunchKernelGGL(k_seg_emit<3>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
+        case 4: hipLaunchKernelGGL(k_seg_emit<4>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
         default: return hipErrorInvalidValue;
         }
+        HIP_TRY(hipGetLastError());
+    }
+    if (op == 2) {
+        ProfScope ps("f64_combine", s);
+        uint32_t gb = nb < 2048 ? nb : 2048;
+        hipLaunchKernelGGL(k_f64_seg_combine, dim3(gb), dim3(BLOCK), 0, s,
+                           lead_seg, lead_part, nchunks, (double *)out_v);
         HIP_TRY(hipGetLastError());
     }
     *h_nout = total;
@@ -1327,6 +1507,7 @@ static hipError_t partition_generic(hipStream_t s, const uint64_t *in_k, const u
                                     uint64_t n, uint32_t nparts, uint64_t *out_k,
                                     uint64_t *out_v, uint64_t *h_counts, DF df, Ws &ws) {
     if (nparts == 0 || nparts > 256) return hipErrorInvalidValue;
+    if (n >= (1ULL << 32)) return hipErrorNotSupported; /* u32 hist/scan limit */
     if (n == 0) {
         for (uint32_t p = 0; p < nparts; ++p) h_counts[p] = 0;
         return hipSuccess;
@@ -1551,6 +1732,8 @@ size_t ws_bytes_for(uint64_t n) {
     /* scan recursion partials: nb/TILE + nb/TILE^2 + ... < nb/2048 */
     b += (((size_t)nb / 2048 + 4096) * 4 + 255) & ~255ULL;
     b += ((size_t)257 * 4 + 255) & ~255ULL;       /* partition starts */
+    b += (size_t)CLEANUP_WL_CAP * 8 + 512;        /* cleanup long-run worklist */
+    b += ((size_t)nb * BLOCK * 12 + 255) & ~255ULL; /* f64 lead partials (seg, OP 2) */
     b += 1 << 20;                                 /* slack */
     return b;
 }
@@ -1599,6 +1782,19 @@ __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
     }
 }
 
+/* u64 total of the per-row match counts — computed BEFORE the u32 prefix
+ * scan so a join whose output exceeds 2^32-1 rows fails loudly instead of
+ * silently wrapping the scan */
+__global__ void k_sum_u32_u64(const uint32_t *a, uint64_t n, unsigned long long *out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    unsigned long long acc = 0;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        acc += a[i];
+    for (int off = 32; off > 0; off >>= 1)
+        acc += (unsigned long long)__shfl_down((unsigned long long)acc, off);
+    if ((threadIdx.x & 63) == 0 && acc) atomicAdd(out, acc);
+}
+
 __global__ void k_join_emit(const int64_t *ak, const int64_t *av, uint64_t na,
                             const int64_t *bv, const uint32_t *scan,
                             const uint32_t *b_lo, int64_t *out_k,
@@ -1625,9 +1821,13 @@ hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint
                        int64_t *out_k, int64_t *out_va, int64_t *out_vb,
                        uint64_t cap, uint64_t *h_nout, Ws &ws) {
     if (na == 0 || nb == 0) { *h_nout = 0; return hipSuccess; }
+    /* per-row counts, b_lo and the prefix scan are u32: side sizes and the
+     * emitted total must each fit below 2^32 (vega_gpu.h) */
+    if (na >= (1ULL << 32) || nb >= (1ULL << 32)) return hipErrorNotSupported;
     uint32_t *counts = (uint32_t *)ws.take((na + 1) * 4);
     uint32_t *b_lo = (uint32_t *)ws.take(na * 4);
-    if (!counts || !b_lo) return hipErrorOutOfMemory;
+    unsigned long long *d_total = (unsigned long long *)ws.take(256);
+    if (!counts || !b_lo || !d_total) return hipErrorOutOfMemory;
     uint32_t nb_grid = nblocks_for(na);
     uint32_t gb = nb_grid < 2048 ? nb_grid : 2048;
     {
@@ -1636,16 +1836,23 @@ hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint
                            mode, counts, b_lo);
         HIP_TRY(hipGetLastError());
     }
+    /* exact u64 total first: a >2^32-row join answers count queries correctly
+     * and REFUSES to emit (the u32 scan would corrupt positions silently) */
+    HIP_TRY(hipMemsetAsync(d_total, 0, 8, s));
+    hipLaunchKernelGGL(k_sum_u32_u64, dim3(gb), dim3(BLOCK), 0, s, counts, na, d_total);
+    HIP_TRY(hipGetLastError());
+    unsigned long long total64 = 0;
+    HIP_TRY(hipMemcpyAsync(&total64, d_total, 8, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    *h_nout = total64;
+    if (!out_k) return hipSuccess; /* count-only query */
+    if (total64 >= (1ULL << 32)) return hipErrorNotSupported;
     HIP_TRY(hipMemsetAsync(counts + na, 0, 4, s));
     {
         Ws w2 = ws;
         HIP_TRY(scan_u32_excl(s, counts, na + 1, w2));
     }
-    uint32_t total = 0;
-    HIP_TRY(hipMemcpyAsync(&total, counts + na, 4, hipMemcpyDeviceToHost, s));
-    HIP_TRY(hipStreamSynchronize(s));
-    *h_nout = total;
-    if (!out_k) return hipSuccess; /* count-only query */
+    uint64_t total = total64;
     if (total > cap) return hipErrorInvalidValue;
     {
         ProfScope ps("join_emit", s);
