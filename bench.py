@@ -107,6 +107,11 @@ def main():
     ap.add_argument("--dtype", choices=["i64", "f64"], default="i64",
                     help="value type for --op reduce (f64 = SURVEY C1's f64-sum variant)")
     ap.add_argument("--zipf-keyspace", type=int, default=100_000_000)
+    ap.add_argument("--force-dist", action="store_true",
+                    help="run the distributed exchange path (RCCL init, "
+                    "counts all-to-all, all-to-all-v) even at WORLD_SIZE=1 — "
+                    "hardware coverage of the collective plumbing on a "
+                    "1-GPU box (launch via torch.distributed.run)")
     args = ap.parse_args()
 
     import torch
@@ -115,7 +120,8 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    if world > 1:
+    dist_on = world > 1 or (args.force_dist and "MASTER_ADDR" in os.environ)
+    if dist_on:
         import torch.distributed as dist
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group("nccl")
@@ -144,12 +150,12 @@ def main():
         v = torch.empty(rows, dtype=torch.int64, device=dev)
         gpu.dev_gen_uniform(k, v, seed=args.seed, key_bits=args.key_bits,
                             start=rank * rows)
-    slack = 1.10 if world > 1 else 1.0
+    slack = 1.10 if dist_on else 1.0
     cap = int(rows * slack) + 1024
     ws = gpu.alloc_ws(cap)
     out_k = torch.empty(cap, dtype=torch.int64, device=dev)
     out_v = torch.empty(cap, dtype=torch.int64, device=dev)
-    if world > 1 or args.op in ("sort", "join"):
+    if dist_on or args.op in ("sort", "join"):
         pk = torch.empty(rows, dtype=torch.int64, device=dev)
         pv = torch.empty(rows, dtype=torch.int64, device=dev)
 
@@ -157,7 +163,7 @@ def main():
 
     def step_reduce(op):
         nonlocal nout
-        if world == 1:
+        if not dist_on:
             nout = gpu.dev_sort_reduce(k, v, op, out_k, out_v, ws)
         else:
             counts = gpu.dev_partition(k, v, world, pk, pv, ws)
@@ -171,7 +177,7 @@ def main():
         # dependency.rs:191-210): local (key,count) aggregate BEFORE the
         # exchange, so Zipf hot keys cross xGMI as one row per rank
         nonlocal nout
-        if world == 1:
+        if not dist_on:
             nout = gpu.dev_sort_reduce(k, v, gpu.OP_COUNT, out_k, out_v, ws)
             return
         nagg = gpu.dev_sort_reduce(k, v, gpu.OP_COUNT, out_k, out_v, ws)
@@ -182,7 +188,7 @@ def main():
 
     def step_sort():
         nonlocal nout
-        if world == 1:
+        if not dist_on:
             out_k[:rows].copy_(k)
             out_v[:rows].copy_(v)
             gpu.dev_sort_pairs(out_k[:rows], out_v[:rows], ws)
@@ -212,7 +218,7 @@ def main():
 
     def step_join():
         nonlocal nout
-        if world == 1:
+        if not dist_on:
             out_k[:rows].copy_(k); out_v[:rows].copy_(v)
             sb_k[:rows].copy_(kb); sb_v[:rows].copy_(vb)
             ta = gpu.dev_group_pairs(out_k[:rows], out_v[:rows], ws)
@@ -247,7 +253,7 @@ def main():
 
     def barrier_sync():
         torch.cuda.synchronize()
-        if world > 1:
+        if dist_on:
             import torch.distributed as dist
             dist.barrier()
             torch.cuda.synchronize()
@@ -257,7 +263,7 @@ def main():
         step()
     barrier_sync()
 
-    profiling = (rank == 0 and world == 1)
+    profiling = (rank == 0 and not dist_on)
     if profiling:
         gpu.prof_enable(True)
 
@@ -268,7 +274,7 @@ def main():
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
-    if world > 1:
+    if dist_on:
         import torch.distributed as dist
         t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
@@ -328,7 +334,7 @@ def main():
         }
         print(json.dumps(out), flush=True)
 
-    if world > 1:
+    if dist_on:
         import torch.distributed as dist
         dist.destroy_process_group()
 

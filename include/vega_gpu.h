@@ -72,8 +72,18 @@ typedef struct vega_ctx vega_ctx_t;
 typedef uint64_t vega_rdd_t;      /* opaque RDD handle, 0 = invalid */
 
 /* ---------------- context ---------------- */
-/* ngpus: number of GPUs this process drives. The rank-per-GPU model uses 1
- * (device = current HIP device); values > 1 are reserved. */
+/* ngpus: number of GPUs this process drives. ngpus == 1 = the rank-per-GPU
+ * model (device = current HIP device). ngpus > 1 = the in-process local-mode
+ * analogue: one process drives all ngpus devices, sharding rows a9-style and
+ * exchanging buckets over RCCL/xGMI (reduce_by_key / group_count /
+ * distinct; other ops on a G>1 context return VEGA_ERR_UNSUPPORTED).
+ *
+ * PER-CALL ROW LIMIT: every op is bounded to n < 2^32 rows per call (u32
+ * histogram/scan plumbing); larger inputs must be sharded (the rank-per-GPU
+ * launcher always does). Calls beyond the limit return
+ * VEGA_ERR_UNSUPPORTED — never a silent wrap. A join whose OUTPUT would
+ * exceed 2^32-1 rows likewise refuses to emit (count queries still return
+ * the exact u64 total). */
 int vega_gpu_init(int ngpus, vega_ctx_t **out);
 int vega_gpu_shutdown(vega_ctx_t *ctx);
 int vega_gpu_synchronize(vega_ctx_t *ctx);
@@ -108,9 +118,34 @@ int vega_gpu_sort_by_key(vega_ctx_t *ctx, vega_rdd_t rdd, vega_rdd_t *out);
 /* inner join via sorted runs (co_grouped_rdd.rs:206-249 + pair_rdd.rs:104-121) */
 int vega_gpu_join(vega_ctx_t *ctx, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
                   vega_rdd_t *out);
-/* distinct (rdd.rs:501-531): keys of rdd deduplicated */
+/* distinct (rdd.rs:501-531): the deduplicated KEY SET (the element column of
+ * this typed engine is the key). Result values are zeroed — use
+ * vega_gpu_group_count for (key, count). */
 int vega_gpu_distinct(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
                       vega_rdd_t *out);
+/* group_by_key (pair_rdd.rs:35-52; aggregator.rs:33-53 Vec-collect): groups
+ * materialized IN THE ENGINE — result is a grouped rdd holding distinct
+ * keys, u64 offsets and the values column in grouped order (value order
+ * within a group = row order; the grouping sort is stable). Collect with
+ * vega_gpu_collect_groups. */
+int vega_gpu_group_by_key(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
+                          vega_rdd_t *out);
+/* cogroup (pair_rdd.rs:123-155 via co_grouped_rdd.rs:206-249): for every key
+ * in EITHER side, the (Vec<V>, Vec<W>) ranges. Single call, host outputs:
+ * keys/offa/lena/offb/lenb sized cap (na+nb always suffices; the needed
+ * count is returned in *nk even on VEGA_ERR_CAP), vala[na] and valb[nb] are
+ * the two value columns in grouped order; key i's A values are
+ * vala[offa[i] .. offa[i]+lena[i]), likewise B. */
+int vega_gpu_cogroup_collect(vega_ctx_t *ctx, vega_rdd_t a, vega_rdd_t b,
+                             int64_t *keys, uint64_t *offa, uint64_t *lena,
+                             uint64_t *offb, uint64_t *lenb, int64_t *vala,
+                             int64_t *valb, uint64_t cap, uint64_t *nk);
+/* intersection / subtract (rdd.rs compositions over CoGroupedRdd): key-set
+ * semantics — distinct keys present in both / in a only. Values zeroed. */
+int vega_gpu_intersection(vega_ctx_t *ctx, vega_rdd_t a, vega_rdd_t b,
+                          uint32_t nparts, vega_rdd_t *out);
+int vega_gpu_subtract(vega_ctx_t *ctx, vega_rdd_t a, vega_rdd_t b,
+                      uint32_t nparts, vega_rdd_t *out);
 /* count_by_value (rdd.rs:449-459 = map(x->(x,1)) + reduce_by_key(+)):
  * counts over the VALUE column; result rows are (value, count) */
 int vega_gpu_count_by_value(vega_ctx_t *ctx, vega_rdd_t rdd, uint32_t nparts,
@@ -130,6 +165,12 @@ int vega_gpu_collect(vega_ctx_t *ctx, vega_rdd_t rdd, int64_t *keys, void *vals,
 /* collect of a join result (K,(V,W)) — pair_rdd.rs:104-121's output shape */
 int vega_gpu_collect_join(vega_ctx_t *ctx, vega_rdd_t rdd, int64_t *keys,
                           int64_t *va, int64_t *vb, uint64_t *n);
+/* collect of a grouped rdd (vega_gpu_group_by_key): keys[nk],
+ * offsets[nk+1] (u64), values[nvals] (int64 or double per the source rdd).
+ * Query sizes with keys == NULL. */
+int vega_gpu_collect_groups(vega_ctx_t *ctx, vega_rdd_t rdd, int64_t *keys,
+                            uint64_t *offsets, void *values, uint64_t *nk,
+                            uint64_t *nvals);
 int vega_gpu_free_rdd(vega_ctx_t *ctx, vega_rdd_t rdd);
 
 /* ---------------- profiling ---------------- */

@@ -31,6 +31,10 @@ struct RddImpl {
     uint64_t alloc_rows = 0;
     uint32_t nparts = 1;
     bool sorted = false;
+    /* grouped rdd (group_by_key): d_k = distinct keys (n of them), d_v =
+     * u64 offsets (n+1), d_v2 = values in grouped order (n2 rows) */
+    bool grouped = false;
+    uint64_t n2 = 0;
     /* multi-GPU (ngpus > 1): per-device shards; d_k/d_v above are device 0's
      * shard so the single-GPU code paths stay untouched for G == 1 */
     std::vector<int64_t *> mk;
@@ -509,22 +513,22 @@ int vega_gpu_sort_by_key(vega_ctx_t *c, vega_rdd_t rdd, vega_rdd_t *out) {
 /* inner join (pair_rdd.rs:104-121 via cogroup co_grouped_rdd.rs:206-249):
  * bring both sides into the GROUPING order (4-5 hash passes instead of the
  * full 8-pass signed sort) and sort-merge with the (h32,key) comparator. */
-int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
-                  vega_rdd_t *out) {
-    if (!c) return VEGA_ERR_INVALID;
-    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED; /* G>1: north-star ops only (this branch) */
-    RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
-    if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
+/* grouped copies of both sides with a CONSISTENT comparator for the
+ * sort-merge kernels: mode 2 ((h32,key) lex) when both group sorts kept the
+ * pinned 4-byte hash order, else both harmonized to the full unsigned-key
+ * order (mode 1). Caller frees *ha / *hb. */
+static int group_two_sides(vega_ctx *c, RddImpl *ra, RddImpl *rb, uint32_t nparts,
+                           vega_rdd_t *ha, vega_rdd_t *hb, RddImpl **sa_out,
+                           RddImpl **sb_out, int *mode_out) {
     uint64_t nmax = ra->n > rb->n ? ra->n : rb->n;
     int rc = ensure_ws(c, nmax);
     if (rc) return rc;
-    /* grouped copies of both sides */
-    vega_rdd_t ha = 0, hb = 0;
     RddImpl *sa, *sb;
-    rc = new_rdd(c, ra->n ? ra->n : 1, 0, nparts, &sa, &ha);
+    *ha = 0; *hb = 0;
+    rc = new_rdd(c, ra->n ? ra->n : 1, 0, nparts, &sa, ha);
     if (rc) return rc;
-    rc = new_rdd(c, rb->n ? rb->n : 1, 0, nparts, &sb, &hb);
-    if (rc) { vega_gpu_free_rdd(c, ha); return rc; }
+    rc = new_rdd(c, rb->n ? rb->n : 1, 0, nparts, &sb, hb);
+    if (rc) { vega_gpu_free_rdd(c, *ha); *ha = 0; return rc; }
     sa->n = ra->n;
     sb->n = rb->n;
     int tag_a = 0, tag_b = 0;
@@ -540,12 +544,9 @@ int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
         Ws wsb(c->ws, c->ws_bytes);
         CTX_TRY(c, group_pairs_inplace(c->stream, sb->d_k, (int64_t *)sb->d_v, sb->n, &tag_b, wsb));
     }
-    /* both sides must share the comparator's order: if either side is not
-     * in the (h32,key) order (narrow keys or fallback), harmonize both to
-     * the full unsigned-key order */
-    int join_mode = 2;
+    int mode = 2;
     if (tag_a != 4 || tag_b != 4) {
-        join_mode = 1;
+        mode = 1;
         const uint64_t *rk, *rv;
         if (sa->n && tag_a == 4) {
             Ws wsa(c->ws, c->ws_bytes);
@@ -566,6 +567,23 @@ int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
             }
         }
     }
+    *sa_out = sa;
+    *sb_out = sb;
+    *mode_out = mode;
+    return VEGA_OK;
+}
+
+int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
+                  vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED; /* G>1: north-star ops only (this branch) */
+    RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
+    if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
+    vega_rdd_t ha = 0, hb = 0;
+    RddImpl *sa, *sb;
+    int join_mode = 2;
+    int rc = group_two_sides(c, ra, rb, nparts, &ha, &hb, &sa, &sb, &join_mode);
+    if (rc) return rc;
     uint64_t total = 0;
     {
         Ws ws(c->ws, c->ws_bytes);
@@ -577,6 +595,13 @@ int vega_gpu_join(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b, uint32_t nparts,
             snprintf(c->err, sizeof c->err, "join count: %s", hipGetErrorString(e));
             return VEGA_ERR_HIP;
         }
+    }
+    if (total >= (1ULL << 32)) { /* u32 emit scan limit: refuse loudly */
+        vega_gpu_free_rdd(c, ha); vega_gpu_free_rdd(c, hb);
+        snprintf(c->err, sizeof c->err,
+                 "join output %llu rows exceeds the 2^32-1 per-call limit",
+                 (unsigned long long)total);
+        return VEGA_ERR_UNSUPPORTED;
     }
     RddImpl *o;
     rc = new_rdd(c, total ? total : 1, 0, nparts, &o, out);
@@ -616,6 +641,245 @@ int vega_gpu_collect_join(vega_ctx_t *c, vega_rdd_t rdd, int64_t *keys,
     }
     CTX_TRY(c, hipStreamSynchronize(c->stream));
     return VEGA_OK;
+}
+
+/* group_by_key (pair_rdd.rs:35-52, aggregator.rs:33-53): groups MATERIALIZED
+ * in the engine — keys + u64 offsets + the values column in grouped order,
+ * all device-resident (replaces the host-side numpy assembly). Value order
+ * within each group is the row order (the grouping sort is stable), matching
+ * the reference's per-partition append order. */
+int vega_gpu_group_by_key(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts,
+                          vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED;
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r) return VEGA_ERR_INVALID;
+    int rc = ensure_ws(c, r->n);
+    if (rc) return rc;
+    RddImpl *o;
+    rc = new_rdd(c, r->n ? r->n : 1, r->vtype, nparts, &o, out);
+    if (rc) return rc;
+    /* d_k: distinct keys; d_v: u64 offsets (reuse the n-row alloc, nk+1 <=
+     * n+1 needs one extra slot) */
+    CTX_TRY(c, hipFree(o->d_v));
+    o->d_v = nullptr;
+    CTX_TRY(c, hipMalloc(&o->d_v, (r->n + 2) * 8));
+    CTX_TRY(c, hipMalloc(&o->d_v2, (r->n ? r->n : 1) * 8));
+    Ws ws(c->ws, c->ws_bytes);
+    const uint64_t *sk, *sv;
+    CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
+                              r->n, 0, nullptr, ws, &sk, &sv));
+    /* values in grouped order */
+    if (r->n)
+        CTX_TRY(c, hipMemcpyAsync(o->d_v2, sv, r->n * 8, hipMemcpyDeviceToDevice, c->stream));
+    /* distinct keys + per-group counts (counts go to a transient) */
+    int64_t *cnt = nullptr;
+    CTX_TRY(c, hipMalloc(&cnt, (r->n ? r->n : 1) * 8));
+    uint64_t nk = 0;
+    {
+        hipError_t e = seg_reduce(c->stream, sk, sv, r->n, VEGA_OP_COUNT,
+                                  (uint64_t *)o->d_k, cnt, &nk, ws);
+        if (e != hipSuccess) {
+            (void)hipFree(cnt);
+            snprintf(c->err, sizeof c->err, "group_by_key: %s", hipGetErrorString(e));
+            return e == hipErrorNotSupported ? VEGA_ERR_UNSUPPORTED : VEGA_ERR_HIP;
+        }
+    }
+    {
+        hipError_t e = counts_to_offsets_u64(c->stream, cnt, nk, (uint64_t *)o->d_v, ws);
+        (void)hipFree(cnt);
+        if (e != hipSuccess) {
+            snprintf(c->err, sizeof c->err, "group offsets: %s", hipGetErrorString(e));
+            return VEGA_ERR_HIP;
+        }
+    }
+    o->n = nk;
+    o->n2 = r->n;
+    o->grouped = true;
+    return VEGA_OK;
+}
+
+/* collect a grouped rdd: keys[nk], offsets[nk+1] (u64), values[nvals].
+ * Query sizes with keys == NULL. */
+int vega_gpu_collect_groups(vega_ctx_t *c, vega_rdd_t rdd, int64_t *keys,
+                            uint64_t *offsets, void *values, uint64_t *nk,
+                            uint64_t *nvals) {
+    RddImpl *r = get_rdd(c, rdd);
+    if (!r || !r->grouped) return VEGA_ERR_INVALID;
+    if (!keys) { *nk = r->n; *nvals = r->n2; return VEGA_OK; }
+    if (*nk < r->n || *nvals < r->n2) return VEGA_ERR_CAP;
+    *nk = r->n;
+    *nvals = r->n2;
+    if (r->n) {
+        CTX_TRY(c, hipMemcpyAsync(keys, r->d_k, r->n * 8, hipMemcpyDeviceToHost, c->stream));
+        CTX_TRY(c, hipMemcpyAsync(offsets, r->d_v, (r->n + 1) * 8, hipMemcpyDeviceToHost, c->stream));
+    }
+    if (r->n2)
+        CTX_TRY(c, hipMemcpyAsync(values, r->d_v2, r->n2 * 8, hipMemcpyDeviceToHost, c->stream));
+    CTX_TRY(c, hipStreamSynchronize(c->stream));
+    return VEGA_OK;
+}
+
+/* cogroup (pair_rdd.rs:123-155 via co_grouped_rdd.rs:206-249): per key
+ * present in EITHER side, the (Vec<V>, Vec<W>) ranges. Single call: caller
+ * provides keys/offa/lena/offb/lenb sized cap >= |keys(a) U keys(b)|
+ * (na+nb always suffices), vala[na], valb[nb]. Outputs: vala/valb are the
+ * two value columns in grouped order; per key i, side A values are
+ * vala[offa[i] .. offa[i]+lena[i]) and likewise for B. */
+int vega_gpu_cogroup_collect(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b,
+                             int64_t *keys, uint64_t *offa, uint64_t *lena,
+                             uint64_t *offb, uint64_t *lenb, int64_t *vala,
+                             int64_t *valb, uint64_t cap, uint64_t *h_nk) {
+    if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED;
+    RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
+    if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
+    vega_rdd_t ha = 0, hb = 0;
+    RddImpl *sa, *sb;
+    int mode = 2;
+    int rc = group_two_sides(c, ra, rb, 1, &ha, &hb, &sa, &sb, &mode);
+    if (rc) return rc;
+    /* distinct keys + counts + offsets per side (device temporaries) */
+    int64_t *ka_u = nullptr, *cnta = nullptr, *kb_u = nullptr, *cntb = nullptr;
+    uint64_t *offa_d = nullptr, *offb_d = nullptr;
+    int64_t *keys_d = nullptr;
+    uint64_t *meta_d = nullptr; /* offa,lena,offb,lenb x cap */
+    uint64_t nka = 0, nkb = 0, nk = 0;
+    hipError_t e = hipSuccess;
+    int ret = VEGA_ERR_HIP;
+    do {
+        if ((e = hipMalloc(&ka_u, (sa->n ? sa->n : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&cnta, (sa->n ? sa->n : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&kb_u, (sb->n ? sb->n : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&cntb, (sb->n ? sb->n : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&offa_d, (sa->n + 2) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&offb_d, (sb->n + 2) * 8)) != hipSuccess) break;
+        {
+            Ws ws(c->ws, c->ws_bytes);
+            if ((e = seg_reduce(c->stream, (const uint64_t *)sa->d_k,
+                                (const uint64_t *)sa->d_v, sa->n, VEGA_OP_COUNT,
+                                (uint64_t *)ka_u, cnta, &nka, ws)) != hipSuccess) break;
+        }
+        {
+            Ws ws(c->ws, c->ws_bytes);
+            if ((e = seg_reduce(c->stream, (const uint64_t *)sb->d_k,
+                                (const uint64_t *)sb->d_v, sb->n, VEGA_OP_COUNT,
+                                (uint64_t *)kb_u, cntb, &nkb, ws)) != hipSuccess) break;
+        }
+        {
+            Ws ws(c->ws, c->ws_bytes);
+            if ((e = counts_to_offsets_u64(c->stream, cnta, nka, offa_d, ws)) != hipSuccess) break;
+            if ((e = counts_to_offsets_u64(c->stream, cntb, nkb, offb_d, ws)) != hipSuccess) break;
+        }
+        uint64_t kcap = nka + nkb;
+        if ((e = hipMalloc(&keys_d, (kcap ? kcap : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&meta_d, (kcap ? kcap : 1) * 4 * 8)) != hipSuccess) break;
+        {
+            Ws ws(c->ws, c->ws_bytes);
+            if ((e = cogroup_index(c->stream, ka_u, nka, offa_d, kb_u, nkb, offb_d,
+                                   mode, keys_d, meta_d, meta_d + kcap,
+                                   meta_d + 2 * kcap, meta_d + 3 * kcap,
+                                   kcap, &nk, ws)) != hipSuccess) break;
+        }
+        *h_nk = nk; /* reported even on CAP so the caller can resize */
+        if (nk > cap) { ret = VEGA_ERR_CAP; e = hipSuccess; break; }
+        if (nk) {
+            if ((e = hipMemcpyAsync(keys, keys_d, nk * 8, hipMemcpyDeviceToHost, c->stream)) != hipSuccess) break;
+            if ((e = hipMemcpyAsync(offa, meta_d, nk * 8, hipMemcpyDeviceToHost, c->stream)) != hipSuccess) break;
+            if ((e = hipMemcpyAsync(lena, meta_d + kcap, nk * 8, hipMemcpyDeviceToHost, c->stream)) != hipSuccess) break;
+            if ((e = hipMemcpyAsync(offb, meta_d + 2 * kcap, nk * 8, hipMemcpyDeviceToHost, c->stream)) != hipSuccess) break;
+            if ((e = hipMemcpyAsync(lenb, meta_d + 3 * kcap, nk * 8, hipMemcpyDeviceToHost, c->stream)) != hipSuccess) break;
+        }
+        if (sa->n && (e = hipMemcpyAsync(vala, sa->d_v, sa->n * 8, hipMemcpyDeviceToHost, c->stream)) != hipSuccess) break;
+        if (sb->n && (e = hipMemcpyAsync(valb, sb->d_v, sb->n * 8, hipMemcpyDeviceToHost, c->stream)) != hipSuccess) break;
+        if ((e = hipStreamSynchronize(c->stream)) != hipSuccess) break;
+        ret = VEGA_OK;
+    } while (0);
+    if (e != hipSuccess) {
+        snprintf(c->err, sizeof c->err, "cogroup: %s", hipGetErrorString(e));
+        if (e == hipErrorNotSupported) ret = VEGA_ERR_UNSUPPORTED;
+    }
+    (void)hipStreamSynchronize(c->stream);
+    if (ka_u) (void)hipFree(ka_u);
+    if (cnta) (void)hipFree(cnta);
+    if (kb_u) (void)hipFree(kb_u);
+    if (cntb) (void)hipFree(cntb);
+    if (offa_d) (void)hipFree(offa_d);
+    if (offb_d) (void)hipFree(offb_d);
+    if (keys_d) (void)hipFree(keys_d);
+    if (meta_d) (void)hipFree(meta_d);
+    vega_gpu_free_rdd(c, ha);
+    vega_gpu_free_rdd(c, hb);
+    return ret;
+}
+
+/* intersection / subtract (rdd.rs set compositions over CoGroupedRdd):
+ * the element column is the key; results are key SETS (values zeroed). */
+static int set_op_common(vega_ctx *c, vega_rdd_t a, vega_rdd_t b, int want,
+                         uint32_t nparts, vega_rdd_t *out) {
+    if (!c) return VEGA_ERR_INVALID;
+    if (c->ngpus > 1) return VEGA_ERR_UNSUPPORTED;
+    RddImpl *ra = get_rdd(c, a), *rb = get_rdd(c, b);
+    if (!ra || !rb || ra->vtype || rb->vtype) return VEGA_ERR_INVALID;
+    vega_rdd_t ha = 0, hb = 0;
+    RddImpl *sa, *sb;
+    int mode = 2;
+    int rc = group_two_sides(c, ra, rb, nparts, &ha, &hb, &sa, &sb, &mode);
+    if (rc) return rc;
+    int64_t *ka_u = nullptr, *cnta = nullptr, *kb_u = nullptr, *cntb = nullptr;
+    uint64_t nka = 0, nkb = 0, nsel = 0;
+    hipError_t e = hipSuccess;
+    int ret = VEGA_ERR_HIP;
+    RddImpl *o = nullptr;
+    do {
+        if ((e = hipMalloc(&ka_u, (sa->n ? sa->n : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&cnta, (sa->n ? sa->n : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&kb_u, (sb->n ? sb->n : 1) * 8)) != hipSuccess) break;
+        if ((e = hipMalloc(&cntb, (sb->n ? sb->n : 1) * 8)) != hipSuccess) break;
+        {
+            Ws ws(c->ws, c->ws_bytes);
+            if ((e = seg_reduce(c->stream, (const uint64_t *)sa->d_k,
+                                (const uint64_t *)sa->d_v, sa->n, VEGA_OP_COUNT,
+                                (uint64_t *)ka_u, cnta, &nka, ws)) != hipSuccess) break;
+        }
+        {
+            Ws ws(c->ws, c->ws_bytes);
+            if ((e = seg_reduce(c->stream, (const uint64_t *)sb->d_k,
+                                (const uint64_t *)sb->d_v, sb->n, VEGA_OP_COUNT,
+                                (uint64_t *)kb_u, cntb, &nkb, ws)) != hipSuccess) break;
+        }
+        int rc2 = new_rdd(c, nka ? nka : 1, 0, nparts, &o, out);
+        if (rc2) { ret = rc2; e = hipSuccess; break; }
+        {
+            Ws ws(c->ws, c->ws_bytes);
+            if ((e = member_select(c->stream, ka_u, nka, kb_u, nkb, mode, want,
+                                   o->d_k, &nsel, ws)) != hipSuccess) break;
+        }
+        o->n = nsel;
+        if (nsel && (e = hipMemsetAsync(o->d_v, 0, nsel * 8, c->stream)) != hipSuccess) break;
+        ret = VEGA_OK;
+    } while (0);
+    if (e != hipSuccess) {
+        snprintf(c->err, sizeof c->err, "set op: %s", hipGetErrorString(e));
+        if (e == hipErrorNotSupported) ret = VEGA_ERR_UNSUPPORTED;
+    }
+    (void)hipStreamSynchronize(c->stream);
+    if (ka_u) (void)hipFree(ka_u);
+    if (cnta) (void)hipFree(cnta);
+    if (kb_u) (void)hipFree(kb_u);
+    if (cntb) (void)hipFree(cntb);
+    vega_gpu_free_rdd(c, ha);
+    vega_gpu_free_rdd(c, hb);
+    return ret;
+}
+
+int vega_gpu_intersection(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b,
+                          uint32_t nparts, vega_rdd_t *out) {
+    return set_op_common(c, a, b, 1, nparts, out);
+}
+int vega_gpu_subtract(vega_ctx_t *c, vega_rdd_t a, vega_rdd_t b,
+                      uint32_t nparts, vega_rdd_t *out) {
+    return set_op_common(c, a, b, 0, nparts, out);
 }
 
 int vega_gpu_map(vega_ctx_t *c, vega_rdd_t rdd, vega_map_op_t op, int64_t p0,

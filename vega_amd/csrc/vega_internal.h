@@ -103,6 +103,25 @@ hipError_t join_sorted(hipStream_t s, const int64_t *ak, const int64_t *av, uint
 hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
                                uint64_t n, int *order_tag, Ws &ws);
 
+/* u64 offsets (nk+1) from i64 group counts (cogroup / group_by_key) */
+hipError_t counts_to_offsets_u64(hipStream_t s, const int64_t *counts, uint64_t nk,
+                                 uint64_t *offsets, Ws &ws);
+
+/* distinct keys of A present (want=1) / absent (want=0) in the sorted
+ * distinct list kb_u — intersection / subtract (rdd.rs set ops) */
+hipError_t member_select(hipStream_t s, const int64_t *ka_u, uint64_t nka,
+                         const int64_t *kb_u, uint64_t nkb, int mode, int want,
+                         int64_t *out_keys, uint64_t *h_nout, Ws &ws);
+
+/* cogroup index: keys + per-key (offa,lena,offb,lenb) over the two distinct
+ * lists and their u64 offsets; h_nk = nka + |B \ A| (co_grouped_rdd.rs
+ * :206-249 output shape) */
+hipError_t cogroup_index(hipStream_t s, const int64_t *ka_u, uint64_t nka,
+                         const uint64_t *offa, const int64_t *kb_u, uint64_t nkb,
+                         const uint64_t *offb, int mode, int64_t *keys,
+                         uint64_t *o_offa, uint64_t *o_lena, uint64_t *o_offb,
+                         uint64_t *o_lenb, uint64_t cap, uint64_t *h_nk, Ws &ws);
+
 size_t ws_bytes_for(uint64_t n);
 
 /* diagnostic phase-cycle buffer (VEGA_PHASE_PROF=1), else nullptr */

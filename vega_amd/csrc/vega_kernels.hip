@@ -1758,6 +1758,194 @@ __device__ __forceinline__ bool join_less(int mode, int64_t a, int64_t b) {
     return (uint64_t)a < (uint64_t)b;
 }
 
+/* ------------------------------------------------------------------ */
+/* cogroup / set-op helpers (pair_rdd.rs:123-155 cogroup output shape;
+ * rdd.rs intersection/subtract are key-set compositions over the same
+ * distinct-key lists). All operate on per-side DISTINCT key lists sorted in
+ * a join_less-consistent order. */
+
+__global__ void k_i64_to_u32(const int64_t *a, uint64_t n, uint32_t *out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        out[i] = (uint32_t)a[i];
+}
+__global__ void k_u32_to_u64(const uint32_t *a, uint64_t n, uint64_t *out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        out[i] = a[i];
+}
+
+/* binary-search each key of `keys` in the sorted list `list`; out[i] = the
+ * matching index, or 0xFFFFFFFF if absent */
+__device__ __forceinline__ bool join_less(int mode, int64_t a, int64_t b);
+__global__ void k_lookup(const int64_t *keys, uint64_t n, const int64_t *list,
+                         uint64_t nl, int mode, uint32_t *out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        int64_t k = keys[i];
+        uint64_t lo = 0, hi = nl;
+        while (lo < hi) {
+            uint64_t m = (lo + hi) >> 1;
+            if (join_less(mode, list[m], k)) lo = m + 1; else hi = m;
+        }
+        out[i] = (lo < nl && list[lo] == k) ? (uint32_t)lo : 0xFFFFFFFFu;
+    }
+}
+
+/* flags[i] = ((lookup[i] != SENT) == want) for the compaction scan */
+__global__ void k_member_flags(const uint32_t *lookup, uint64_t n, int want,
+                               uint32_t *flags) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        flags[i] = ((lookup[i] != 0xFFFFFFFFu) ? 1 : 0) == want ? 1u : 0u;
+}
+
+/* emit keys whose scanned flag advanced (flags = exclusive scan, n+1) */
+__global__ void k_compact_keys(const int64_t *keys, uint64_t n, const uint32_t *scan,
+                               int64_t *out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        if (scan[i + 1] > scan[i]) out[scan[i]] = keys[i];
+}
+
+/* cogroup emit, A-derived rows: every distinct key of A, with its B ranges
+ * where matched */
+__global__ void k_cogroup_emit_a(const int64_t *ka_u, uint64_t nka,
+                                 const uint64_t *offa, const uint64_t *offb,
+                                 const uint32_t *bidx, int64_t *keys,
+                                 uint64_t *o_offa, uint64_t *o_lena,
+                                 uint64_t *o_offb, uint64_t *o_lenb) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nka; i += stride) {
+        keys[i] = ka_u[i];
+        o_offa[i] = offa[i];
+        o_lena[i] = offa[i + 1] - offa[i];
+        uint32_t j = bidx[i];
+        if (j != 0xFFFFFFFFu) {
+            o_offb[i] = offb[j];
+            o_lenb[i] = offb[j + 1] - offb[j];
+        } else {
+            o_offb[i] = 0;
+            o_lenb[i] = 0;
+        }
+    }
+}
+
+/* cogroup emit, B-only rows appended after the nka A rows (scan = exclusive
+ * scan of the not-in-A flags over nkb+1) */
+__global__ void k_cogroup_emit_b(const int64_t *kb_u, uint64_t nkb,
+                                 const uint64_t *offb, const uint32_t *scan,
+                                 uint64_t base, int64_t *keys,
+                                 uint64_t *o_offa, uint64_t *o_lena,
+                                 uint64_t *o_offb, uint64_t *o_lenb) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; j < nkb; j += stride) {
+        if (scan[j + 1] == scan[j]) continue; /* present in A: already emitted */
+        uint64_t pos = base + scan[j];
+        keys[pos] = kb_u[j];
+        o_offa[pos] = 0;
+        o_lena[pos] = 0;
+        o_offb[pos] = offb[j];
+        o_lenb[pos] = offb[j + 1] - offb[j];
+    }
+}
+
+/* u64 offsets (nk+1) from i64 group counts — totals are < 2^32 by the
+ * per-call row guard, so the scan itself runs in u32 */
+hipError_t counts_to_offsets_u64(hipStream_t s, const int64_t *counts, uint64_t nk,
+                                 uint64_t *offsets, Ws &ws) {
+    Ws w2 = ws;
+    uint32_t *tmp = (uint32_t *)w2.take((nk + 1) * 4);
+    if (!tmp) return hipErrorOutOfMemory;
+    uint32_t gb = (uint32_t)((nk / BLOCK) + 1);
+    if (gb > 2048) gb = 2048;
+    hipLaunchKernelGGL(k_i64_to_u32, dim3(gb), dim3(BLOCK), 0, s, counts, nk, tmp);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipMemsetAsync(tmp + nk, 0, 4, s));
+    HIP_TRY(scan_u32_excl(s, tmp, nk + 1, w2));
+    hipLaunchKernelGGL(k_u32_to_u64, dim3(gb), dim3(BLOCK), 0, s, tmp, nk + 1, offsets);
+    return hipGetLastError();
+}
+
+/* membership select: out = distinct keys of A (ka_u) that are (want=1) /
+ * are not (want=0) present in kb_u. h_nout = emitted count. */
+hipError_t member_select(hipStream_t s, const int64_t *ka_u, uint64_t nka,
+                         const int64_t *kb_u, uint64_t nkb, int mode, int want,
+                         int64_t *out_keys, uint64_t *h_nout, Ws &ws) {
+    if (nka == 0) { *h_nout = 0; return hipSuccess; }
+    Ws w2 = ws;
+    uint32_t *bidx = (uint32_t *)w2.take(nka * 4);
+    uint32_t *flags = (uint32_t *)w2.take((nka + 1) * 4);
+    if (!bidx || !flags) return hipErrorOutOfMemory;
+    uint32_t gb = (uint32_t)((nka / BLOCK) + 1);
+    if (gb > 2048) gb = 2048;
+    hipLaunchKernelGGL(k_lookup, dim3(gb), dim3(BLOCK), 0, s, ka_u, nka, kb_u, nkb,
+                       mode, bidx);
+    HIP_TRY(hipGetLastError());
+    hipLaunchKernelGGL(k_member_flags, dim3(gb), dim3(BLOCK), 0, s, bidx, nka, want, flags);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipMemsetAsync(flags + nka, 0, 4, s));
+    HIP_TRY(scan_u32_excl(s, flags, nka + 1, w2));
+    uint32_t total = 0;
+    HIP_TRY(hipMemcpyAsync(&total, flags + nka, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    hipLaunchKernelGGL(k_compact_keys, dim3(gb), dim3(BLOCK), 0, s, ka_u, nka, flags,
+                       out_keys);
+    HIP_TRY(hipGetLastError());
+    *h_nout = total;
+    return hipSuccess;
+}
+
+/* cogroup index build: from the two distinct lists + their u64 offsets,
+ * produce keys + per-key (offa,lena,offb,lenb). h_nk = nka + |B \ A|. */
+hipError_t cogroup_index(hipStream_t s, const int64_t *ka_u, uint64_t nka,
+                         const uint64_t *offa, const int64_t *kb_u, uint64_t nkb,
+                         const uint64_t *offb, int mode, int64_t *keys,
+                         uint64_t *o_offa, uint64_t *o_lena, uint64_t *o_offb,
+                         uint64_t *o_lenb, uint64_t cap, uint64_t *h_nk, Ws &ws) {
+    Ws w2 = ws;
+    uint32_t *bidx = (uint32_t *)w2.take((nka ? nka : 1) * 4);
+    uint32_t *bscan = (uint32_t *)w2.take((nkb + 1) * 4);
+    if (!bidx || !bscan) return hipErrorOutOfMemory;
+    uint32_t gba = (uint32_t)((nka / BLOCK) + 1);
+    if (gba > 2048) gba = 2048;
+    uint32_t gbb = (uint32_t)((nkb / BLOCK) + 1);
+    if (gbb > 2048) gbb = 2048;
+    if (nka) {
+        hipLaunchKernelGGL(k_lookup, dim3(gba), dim3(BLOCK), 0, s, ka_u, nka, kb_u, nkb,
+                           mode, bidx);
+        HIP_TRY(hipGetLastError());
+    }
+    uint32_t extra = 0;
+    if (nkb) {
+        uint32_t *aidx = bscan; /* reuse: lookup b->a, then flags in place */
+        hipLaunchKernelGGL(k_lookup, dim3(gbb), dim3(BLOCK), 0, s, kb_u, nkb, ka_u, nka,
+                           mode, aidx);
+        HIP_TRY(hipGetLastError());
+        hipLaunchKernelGGL(k_member_flags, dim3(gbb), dim3(BLOCK), 0, s, aidx, nkb,
+                           0, bscan);
+        HIP_TRY(hipGetLastError());
+        HIP_TRY(hipMemsetAsync(bscan + nkb, 0, 4, s));
+        HIP_TRY(scan_u32_excl(s, bscan, nkb + 1, w2));
+        HIP_TRY(hipMemcpyAsync(&extra, bscan + nkb, 4, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+    }
+    uint64_t nk = nka + extra;
+    *h_nk = nk;
+    if (nk > cap) return hipErrorInvalidValue;
+    if (nka) {
+        hipLaunchKernelGGL(k_cogroup_emit_a, dim3(gba), dim3(BLOCK), 0, s, ka_u, nka,
+                           offa, offb, bidx, keys, o_offa, o_lena, o_offb, o_lenb);
+        HIP_TRY(hipGetLastError());
+    }
+    if (nkb && extra) {
+        hipLaunchKernelGGL(k_cogroup_emit_b, dim3(gbb), dim3(BLOCK), 0, s, kb_u, nkb,
+                           offb, bscan, nka, keys, o_offa, o_lena, o_offb, o_lenb);
+        HIP_TRY(hipGetLastError());
+    }
+    return hipSuccess;
+}
+
 __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
                              uint64_t nb, int mode, uint32_t *counts,
                              uint32_t *b_lo) {

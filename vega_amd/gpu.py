@@ -183,16 +183,72 @@ class Rdd:
         return Rdd(self.ctx, out.value, np.int64)
 
     def group_by_key(self):
-        """Full groups (pair_rdd.rs:35-52): (keys, offsets, values) with
-        values in row order per group (via the stable sort_by_key)."""
-        srt = self.sort_by_key()
-        k, v = srt.collect()
-        srt.free()
-        if len(k) == 0:
-            return k, np.zeros(1, dtype=np.int64), v
-        heads = np.flatnonzero(np.concatenate([[True], k[1:] != k[:-1]]))
-        offsets = np.concatenate([heads, [len(k)]]).astype(np.int64)
-        return k[heads], offsets, v
+        """Full groups (pair_rdd.rs:35-52, aggregator.rs:33-53) materialized
+        IN THE ENGINE (vega_gpu_group_by_key): returns (keys, offsets,
+        values) — values in grouped order, value order within a group = row
+        order (stable grouping sort)."""
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_group_by_key(self.ctx._c, ctypes.c_uint64(self.h),
+                                           ctypes.c_uint32(256), ctypes.byref(out)),
+               "group_by_key", self.ctx._c)
+        g = Rdd(self.ctx, out.value, self.vdtype)
+        nk = ctypes.c_uint64(0)
+        nvals = ctypes.c_uint64(0)
+        _check(lib().vega_gpu_collect_groups(self.ctx._c, ctypes.c_uint64(g.h),
+                                             None, None, None,
+                                             ctypes.byref(nk), ctypes.byref(nvals)),
+               "collect_groups size", self.ctx._c)
+        keys = np.empty(nk.value, dtype=np.int64)
+        offsets = np.empty(nk.value + 1, dtype=np.uint64)
+        values = np.empty(nvals.value, dtype=self.vdtype)
+        _check(lib().vega_gpu_collect_groups(self.ctx._c, ctypes.c_uint64(g.h),
+                                             _pp(keys), _pp(offsets), _pp(values),
+                                             ctypes.byref(nk), ctypes.byref(nvals)),
+               "collect_groups", self.ctx._c)
+        g.free()
+        return keys, offsets.astype(np.int64), values
+
+    def cogroup(self, other):
+        """cogroup (pair_rdd.rs:123-155): for every key in either side the
+        (Vec<V>, Vec<W>) ranges. Returns (keys, offa, lena, offb, lenb,
+        vala, valb)."""
+        na, nb = self.count(), other.count()
+        cap = na + nb
+        keys = np.empty(max(cap, 1), dtype=np.int64)
+        offa = np.empty(max(cap, 1), dtype=np.uint64)
+        lena = np.empty(max(cap, 1), dtype=np.uint64)
+        offb = np.empty(max(cap, 1), dtype=np.uint64)
+        lenb = np.empty(max(cap, 1), dtype=np.uint64)
+        vala = np.empty(max(na, 1), dtype=np.int64)
+        valb = np.empty(max(nb, 1), dtype=np.int64)
+        nk = ctypes.c_uint64(0)
+        _check(lib().vega_gpu_cogroup_collect(
+            self.ctx._c, ctypes.c_uint64(self.h), ctypes.c_uint64(other.h),
+            _pp(keys), _pp(offa), _pp(lena), _pp(offb), _pp(lenb),
+            _pp(vala), _pp(valb), ctypes.c_uint64(cap), ctypes.byref(nk)),
+            "cogroup", self.ctx._c)
+        n = nk.value
+        return (keys[:n], offa[:n].astype(np.int64), lena[:n].astype(np.int64),
+                offb[:n].astype(np.int64), lenb[:n].astype(np.int64),
+                vala[:na], valb[:nb])
+
+    def intersection(self, other, nparts=256):
+        """distinct keys present in both sides (rdd.rs set semantics)"""
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_intersection(self.ctx._c, ctypes.c_uint64(self.h),
+                                           ctypes.c_uint64(other.h),
+                                           ctypes.c_uint32(nparts), ctypes.byref(out)),
+               "intersection", self.ctx._c)
+        return Rdd(self.ctx, out.value, np.int64)
+
+    def subtract(self, other, nparts=256):
+        """distinct keys of self absent from other"""
+        out = ctypes.c_uint64()
+        _check(lib().vega_gpu_subtract(self.ctx._c, ctypes.c_uint64(self.h),
+                                       ctypes.c_uint64(other.h),
+                                       ctypes.c_uint32(nparts), ctypes.byref(out)),
+               "subtract", self.ctx._c)
+        return Rdd(self.ctx, out.value, np.int64)
 
     def group_count(self, nparts=256):
         out = ctypes.c_uint64()

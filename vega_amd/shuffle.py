@@ -43,24 +43,46 @@ def partition_cpu(k, v, nparts):
 def choose_splitters(keys, world, group=None, samples=4096):
     """Range-partitioner boundaries for sort_by_key's exchange (the analogue
     of Spark's RangePartitioner sampling; reference has only take_ordered,
-    rdd.rs:1106-1153): every rank contributes `samples` strided key samples,
-    ranks all-gather them, and the world-1 quantiles become the splitters.
-    Returns an int64 tensor of world-1 ascending splitters on keys.device."""
+    rdd.rs:1106-1153): every rank contributes up to `samples` strided key
+    samples plus its row count; quantiles are taken over the VALID samples
+    only, each weighted by the rows it represents (n_r / s_r), so short or
+    empty ranks neither pad nor skew the boundaries. Returns an int64 tensor
+    of world-1 ascending splitters on keys.device."""
     n = keys.numel()
-    if n > 0:
-        idx = torch.linspace(0, n - 1, steps=min(samples, n), dtype=torch.int64,
-                             device=keys.device)
+    dev = keys.device
+    s_local = min(samples, n)
+    if s_local > 0:
+        idx = torch.linspace(0, n - 1, steps=s_local, dtype=torch.int64, device=dev)
         local = keys[idx]
-        if local.numel() < samples:  # pad to fixed size for all_gather
-            pad = local[-1].repeat(samples - local.numel())
-            local = torch.cat([local, pad])
+        if s_local < samples:  # pad to fixed size for all_gather (dropped below)
+            local = torch.cat([local, local.new_zeros(samples - s_local)])
     else:
-        local = torch.zeros(samples, dtype=torch.int64, device=keys.device)
+        local = torch.zeros(samples, dtype=torch.int64, device=dev)
+    meta = torch.tensor([n], dtype=torch.int64, device=dev)
     gathered = [torch.empty_like(local) for _ in range(world)]
+    gmeta = [torch.empty_like(meta) for _ in range(world)]
     dist.all_gather(gathered, local, group=group)
-    allk = torch.cat(gathered).sort().values
-    pos = [(i * allk.numel()) // world for i in range(1, world)]
-    return allk[torch.tensor(pos, dtype=torch.int64, device=keys.device)].contiguous()
+    dist.all_gather(gmeta, meta, group=group)
+    vals, wts = [], []
+    for r in range(world):
+        n_r = int(gmeta[r].item())
+        s_r = min(samples, n_r)
+        if s_r == 0:
+            continue
+        vals.append(gathered[r][:s_r])
+        wts.append(torch.full((s_r,), n_r / s_r, dtype=torch.float64, device=dev))
+    if not vals:
+        return torch.zeros(world - 1, dtype=torch.int64, device=dev)
+    allk = torch.cat(vals)
+    allw = torch.cat(wts)
+    order = allk.argsort()
+    allk = allk[order]
+    cumw = allw[order].cumsum(0)
+    total = float(cumw[-1].item())
+    targets = torch.tensor([i * total / world for i in range(1, world)],
+                           dtype=torch.float64, device=dev)
+    pos = torch.searchsorted(cumw, targets).clamp_(0, allk.numel() - 1)
+    return allk[pos].contiguous()
 
 
 def partition_range_cpu(k, v, splitters):
