@@ -216,34 +216,35 @@ def main():
         sb_k = torch.empty(cap, dtype=torch.int64, device=dev)
         sb_v = torch.empty(cap, dtype=torch.int64, device=dev)
 
+    def _join_grouped_sides(ak_t, av_t, bk_t, bv_t):
+        # group both sides; tag 4 = (h32,key) lex, tag 0 = full unsigned-key
+        # order (narrow keys / fallback — ALREADY sorted, no extra pass).
+        # Mixed tags: only the tag-4 side needs harmonizing to key order.
+        ta = gpu.dev_group_pairs(ak_t, av_t, ws)
+        tb = gpu.dev_group_pairs(bk_t, bv_t, ws)
+        if ta == 4 and tb == 4:
+            return gpu.dev_join_grouped(ak_t, av_t, bk_t, bv_t, 2, jk, jva, jvb, ws)
+        if ta == 0 and tb == 0:  # both already in full unsigned-key order
+            return gpu.dev_join_grouped(ak_t, av_t, bk_t, bv_t, 1, jk, jva, jvb, ws)
+        # mixed tags (rare: one narrow-key side): harmonize BOTH to the
+        # signed order dev_sort_pairs produces and merge with mode 0
+        gpu.dev_sort_pairs(ak_t, av_t, ws)
+        gpu.dev_sort_pairs(bk_t, bv_t, ws)
+        return gpu.dev_join_grouped(ak_t, av_t, bk_t, bv_t, 0, jk, jva, jvb, ws)
+
     def step_join():
         nonlocal nout
         if not dist_on:
             out_k[:rows].copy_(k); out_v[:rows].copy_(v)
             sb_k[:rows].copy_(kb); sb_v[:rows].copy_(vb)
-            ta = gpu.dev_group_pairs(out_k[:rows], out_v[:rows], ws)
-            tb = gpu.dev_group_pairs(sb_k[:rows], sb_v[:rows], ws)
-            if ta == 4 and tb == 4:
-                nout = gpu.dev_join_grouped(out_k[:rows], out_v[:rows],
-                                            sb_k[:rows], sb_v[:rows], 2, jk, jva, jvb, ws)
-            else:  # harmonize to the unsigned-key order
-                gpu.dev_sort_pairs(out_k[:rows], out_v[:rows], ws)
-                gpu.dev_sort_pairs(sb_k[:rows], sb_v[:rows], ws)
-                nout = gpu.dev_join_sorted(out_k[:rows], out_v[:rows],
-                                           sb_k[:rows], sb_v[:rows], jk, jva, jvb, ws)
+            nout = _join_grouped_sides(out_k[:rows], out_v[:rows],
+                                       sb_k[:rows], sb_v[:rows])
         else:
             ca = gpu.dev_partition(k, v, world, pk, pv, ws)
             rak, rav = shuffle.all_to_all_kv(pk, pv, ca.astype(np.int64).tolist())
             cb = gpu.dev_partition(kb, vb, world, pk, pv, ws)
             rbk, rbv = shuffle.all_to_all_kv(pk, pv, cb.astype(np.int64).tolist())
-            ta = gpu.dev_group_pairs(rak, rav, ws)
-            tb = gpu.dev_group_pairs(rbk, rbv, ws)
-            if ta == 4 and tb == 4:
-                nout = gpu.dev_join_grouped(rak, rav, rbk, rbv, 2, jk, jva, jvb, ws)
-            else:
-                gpu.dev_sort_pairs(rak, rav, ws)
-                gpu.dev_sort_pairs(rbk, rbv, ws)
-                nout = gpu.dev_join_sorted(rak, rav, rbk, rbv, jk, jva, jvb, ws)
+            nout = _join_grouped_sides(rak, rav, rbk, rbv)
 
     step = {"reduce": lambda: step_reduce(
                 gpu.OP_SUM_F64 if args.dtype == "f64" else gpu.OP_SUM_I64),

@@ -68,3 +68,23 @@ def test_join_c4_shape_scaled(ctx):
     exp = sorted(zip(ok.tolist(), ova.tolist(), ovb.tolist()))
     assert got == exp
     a.free(); b.free(); j.free()
+
+
+def test_join_mixed_order_tags(ctx):
+    """one narrow-key side (grouping falls back to the skipped key sort, tag
+    0) joined with a wide-key side (pinned hash order, tag 4): the engine
+    must harmonize both to one comparator before the merge"""
+    na, nb = 300_000, 200_000
+    ak, av = datagen.uniform_pairs(51, na, key_bits=8)     # narrow: tag 0
+    bk_w, bv = datagen.uniform_pairs(52, nb, key_bits=64)  # wide (negatives): tag 4
+    # overlap: map half of b's keys into a's narrow range
+    bk = np.where(np.arange(nb) % 2 == 0, bk_w & 0xFF, bk_w)
+    a = ctx.make_rdd(ak, av)
+    b = ctx.make_rdd(bk, bv)
+    j = a.join(b)
+    k, va, vb = j.collect_join()
+    got = sorted(zip(k.tolist(), va.tolist(), vb.tolist()))
+    ok, ova, ovb = oc.join_i64(ak, av, bk, bv, 4, 4)
+    exp = sorted(zip(ok.tolist(), ova.tolist(), ovb.tolist()))
+    assert got == exp
+    a.free(); b.free(); j.free()

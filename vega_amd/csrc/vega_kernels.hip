@@ -327,8 +327,27 @@ __global__ void k_hist8t(const uint64_t *keys, uint64_t n, uint32_t *h8) {
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         uint64_t k = keys[i];
         if (HASHSRC) k = vega_hash_u64(k);
+        /* uniform-byte fast path: narrow key ranges make whole byte
+         * positions constant across the wave (e.g. [0,5e8) keys: bytes 4-7
+         * all zero), and 64 lanes hitting ONE LDS counter serialize 64-way.
+         * One compare-ballot per byte detects it; the leader then adds the
+         * lane count once. Random bytes fall through to per-lane atomics
+         * (full ballot-dedup was measured 7x slower there — DESIGN.md). */
+        uint64_t act = __ballot(true);
+        int leader = (int)__ffsll((unsigned long long)act) - 1;
+        int lane = threadIdx.x & 63;
+        int nact = __popcll(act);
 #pragma unroll
-        for (int b = 0; b < 8; ++b) atomicAdd(&h[w][b][(k >> (8 * b)) & 0xFF], 1u);
+        for (int b = 0; b < 8; ++b) {
+            uint32_t d = (uint32_t)(k >> (8 * b)) & 0xFF;
+            uint32_t d0 = (uint32_t)__shfl((int)d, leader);
+            uint64_t same = __ballot(d == d0);
+            if (same == act) {
+                if (lane == leader) atomicAdd(&h[w][b][d], (uint32_t)nact);
+            } else {
+                atomicAdd(&h[w][b][d], 1u);
+            }
+        }
     }
     __syncthreads();
     const uint32_t *hf = (const uint32_t *)h;
@@ -609,19 +628,25 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 /* bounded spin: a lost predecessor can never wedge the GPU —
                  * set the abort flag, bail, and let the host fail loudly */
                 if (++spins > (1u << 26)) { *d_abort = 1; break; }
-                /* probe up to 4 predecessors with independent loads */
-                unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
-                int navail = (j >= 3) ? 4 : (int)(j + 1);
-                d0 = __hip_atomic_load(col + j * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 1) d1 = __hip_atomic_load(col + (j - 1) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 2) d2 = __hip_atomic_load(col + (j - 2) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 3) d3 = __hip_atomic_load(col + (j - 3) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                /* probe up to 16 predecessors with independent loads in
+                 * flight together (one L2 round trip per iteration; the walk
+                 * length tracks the ~512 concurrently-resident tiles, so
+                 * deeper probing divides the serial-iteration count — phase
+                 * profiling put 62% of wave cycles here at depth 4) */
+                constexpr int PD = 16;
+                unsigned long long dd16[PD];
+                int navail = (j >= PD - 1) ? PD : (int)(j + 1);
+#pragma unroll
+                for (int q = 0; q < PD; ++q)
+                    if (q < navail)
+                        dd16[q] = __hip_atomic_load(col + (j - q) * 256,
+                                                    __ATOMIC_RELAXED,
+                                                    __HIP_MEMORY_SCOPE_AGENT);
                 bool done = false, stall = false;
-                unsigned long long dd4[4] = {d0, d1, d2, d3};
                 for (int q = 0; q < navail; ++q) {
-                    unsigned long long st = dd4[q] >> 62;
-                    if (st == 2) { excl_tiles += dd4[q] & OSW_CNT_MASK; done = true; break; }
-                    if (st == 1) { excl_tiles += dd4[q] & OSW_CNT_MASK; j--; continue; }
+                    unsigned long long st = dd16[q] >> 62;
+                    if (st == 2) { excl_tiles += dd16[q] & OSW_CNT_MASK; done = true; break; }
+                    if (st == 1) { excl_tiles += dd16[q] & OSW_CNT_MASK; j--; continue; }
                     stall = true;
                     break;
                 }
