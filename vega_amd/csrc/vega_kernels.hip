@@ -333,20 +333,25 @@ __global__ void k_hist8t(const uint64_t *keys, uint64_t n, uint32_t *h8) {
          * One compare-ballot per byte detects it; the leader then adds the
          * lane count once. Random bytes fall through to per-lane atomics
          * (full ballot-dedup was measured 7x slower there — DESIGN.md). */
-        uint64_t act = __ballot(true);
-        int leader = (int)__ffsll((unsigned long long)act) - 1;
-        int lane = threadIdx.x & 63;
-        int nact = __popcll(act);
+        if (!HASHSRC) { /* raw keys: narrow ranges make bytes wave-constant */
+            uint64_t act = __ballot(true);
+            int leader = (int)__ffsll((unsigned long long)act) - 1;
+            int lane = threadIdx.x & 63;
+            int nact = __popcll(act);
 #pragma unroll
-        for (int b = 0; b < 8; ++b) {
-            uint32_t d = (uint32_t)(k >> (8 * b)) & 0xFF;
-            uint32_t d0 = (uint32_t)__shfl((int)d, leader);
-            uint64_t same = __ballot(d == d0);
-            if (same == act) {
-                if (lane == leader) atomicAdd(&h[w][b][d], (uint32_t)nact);
-            } else {
-                atomicAdd(&h[w][b][d], 1u);
+            for (int b = 0; b < 8; ++b) {
+                uint32_t d = (uint32_t)(k >> (8 * b)) & 0xFF;
+                uint32_t d0 = (uint32_t)__shfl((int)d, leader);
+                uint64_t same = __ballot(d == d0);
+                if (same == act) {
+                    if (lane == leader) atomicAdd(&h[w][b][d], (uint32_t)nact);
+                } else {
+                    atomicAdd(&h[w][b][d], 1u);
+                }
             }
+        } else { /* hash bytes are uniform-random: the check never pays */
+#pragma unroll
+            for (int b = 0; b < 8; ++b) atomicAdd(&h[w][b][(k >> (8 * b)) & 0xFF], 1u);
         }
     }
     __syncthreads();
@@ -491,7 +496,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
     uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, int *d_abort,
-    unsigned long long *phc, DF df) {
+    unsigned long long *ff, unsigned long long *phc, DF df) {
     /* phc (diagnostic builds, VEGA_PHASE_PROF=1): per-phase shader-cycle
      * sums, one sample per wave — phases: 0 prefetch, 1 rank, 2 publish+
      * starts, 3 lookback, 4 reorder, 5 writeout (s_memtime; the microarch
@@ -622,38 +627,55 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         unsigned long long excl_tiles = 0;
         if (vb > 0) {
             gdesc_t *col = (gdesc_t *)(desc + t);
+            /* FAST-FORWARD: ff[t] holds (m << 42 | count) = the largest
+             * published inclusive prefix for this digit — count covers tiles
+             * [0, m). Snapshot it and walk only (m .. vb-1], not all ~512
+             * resident predecessors (phase profiling: the full walk was 62%
+             * of all wave cycles). The slot is one relaxed agent-scope
+             * 8-byte atomicMax granule — monotone because inclusive counts
+             * are nondecreasing in tile id. */
+            gdesc_t *slot = (gdesc_t *)(ff + t);
+            unsigned long long snap =
+                __hip_atomic_load(slot, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            int64_t m = (int64_t)(snap >> 42);
+            unsigned long long base = snap & ((1ULL << 42) - 1);
+            if (m > (int64_t)vb) { m = 0; base = 0; } /* snapshot from a later tile: unusable */
+            bool hit_inc = false;
             int64_t j = (int64_t)vb - 1;
             uint32_t spins = 0;
-            while (j >= 0) {
+            while (j >= m) {
                 /* bounded spin: a lost predecessor can never wedge the GPU —
                  * set the abort flag, bail, and let the host fail loudly */
                 if (++spins > (1u << 26)) { *d_abort = 1; break; }
-                /* probe up to 16 predecessors with independent loads in
-                 * flight together (one L2 round trip per iteration; the walk
-                 * length tracks the ~512 concurrently-resident tiles, so
-                 * deeper probing divides the serial-iteration count — phase
-                 * profiling put 62% of wave cycles here at depth 4) */
-                constexpr int PD = 16;
-                unsigned long long dd16[PD];
-                int navail = (j >= PD - 1) ? PD : (int)(j + 1);
-#pragma unroll
-                for (int q = 0; q < PD; ++q)
-                    if (q < navail)
-                        dd16[q] = __hip_atomic_load(col + (j - q) * 256,
-                                                    __ATOMIC_RELAXED,
-                                                    __HIP_MEMORY_SCOPE_AGENT);
+                /* probe up to 4 predecessors with independent loads */
+                unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
+                int navail = (j - m >= 3) ? 4 : (int)(j - m + 1);
+                d0 = __hip_atomic_load(col + j * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 1) d1 = __hip_atomic_load(col + (j - 1) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 2) d2 = __hip_atomic_load(col + (j - 2) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 3) d3 = __hip_atomic_load(col + (j - 3) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 bool done = false, stall = false;
+                unsigned long long dd4[4] = {d0, d1, d2, d3};
                 for (int q = 0; q < navail; ++q) {
-                    unsigned long long st = dd16[q] >> 62;
-                    if (st == 2) { excl_tiles += dd16[q] & OSW_CNT_MASK; done = true; break; }
-                    if (st == 1) { excl_tiles += dd16[q] & OSW_CNT_MASK; j--; continue; }
+                    unsigned long long st = dd4[q] >> 62;
+                    if (st == 2) { /* INC covers tiles [0, j]: base unused */
+                        excl_tiles += dd4[q] & OSW_CNT_MASK;
+                        hit_inc = true;
+                        done = true;
+                        break;
+                    }
+                    if (st == 1) { excl_tiles += dd4[q] & OSW_CNT_MASK; j--; continue; }
                     stall = true;
                     break;
                 }
                 if (done) break;
                 if (stall) __builtin_amdgcn_s_sleep(1);
             }
+            if (!hit_inc) excl_tiles += base;
         }
+        __hip_atomic_fetch_max((gdesc_t *)(ff + t),
+                               ((unsigned long long)(vb + 1) << 42) | (excl_tiles + cnt),
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            (excl_tiles + cnt) | OSW_ST_INC,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -727,29 +749,31 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    uint64_t n, const uint32_t *gbase_d,
                                    unsigned long long *desc, uint32_t *ticket,
                                    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out,
-                                   int *d_abort, bool has_vals, bool in_pk, bool out_pk,
+                                   int *d_abort, unsigned long long *ff,
+                                   bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     unsigned long long *phc = phase_prof_buf();
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
+    HIP_TRY(hipMemsetAsync(ff, 0, 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
         hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, phc, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
     } else if (!in_pk && !out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, phc, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, df);
     } else if (!in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, phc, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
     } else if (in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, phc, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
     } else {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, phc, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, df);
     }
     return hipGetLastError();
 }
@@ -832,7 +856,8 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_abort = (int *)ws.take(256);
-    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort)
+    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 8);
+    if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort || !ff_d)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
 
@@ -880,11 +905,11 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         if (p == 7 && signed_order) {
             RadixDigitTopSigned df{56};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, d_abort, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, ff_d, has_vals, in_pk, out_pk, df, "radix_scatter"));
         } else {
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, d_abort, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, ff_d, has_vals, in_pk, out_pk, df, "radix_scatter"));
         }
         cur = dk;
     }
@@ -1066,11 +1091,12 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
     int *d_abort = (int *)ws.take(256);
+    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 8);
     uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
     unsigned long long *wl = (unsigned long long *)ws.take(CLEANUP_WL_CAP * 8);
     uint32_t *wl_count = (uint32_t *)ws.take(256);
     if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_err || !d_abort ||
-        !h32buf || !wl || !wl_count)
+        !ff_d || !h32buf || !wl || !wl_count)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
     /* strict order (full (h32,key) lex within runs) only when a caller will
@@ -1128,7 +1154,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur, i == 0 ? in_v : nullptr,
                                      n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, d_abort, true, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, ff_d, true, in_pk, out_pk, df, "radix_scatter"));
             cur = dk;
         }
         *rk = cur;
@@ -1170,7 +1196,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur_k, i == 0 ? in_v : nullptr,
                                  n, gbase_d + i * 256, desc, ticket,
                                  dk, dv, out_pk ? nullptr : h32buf,
-                                 d_abort, true, in_pk, out_pk, df, "radix_scatter"));
+                                 d_abort, ff_d, true, in_pk, out_pk, df, "radix_scatter"));
         cur_k = dk;
         cur_v = out_pk ? nullptr : dk + n;
     }
