@@ -61,6 +61,7 @@ def main():
     print(f"rows={n} reps={args.reps} op={args.op} total wave-cycles={total}")
     for i, name in enumerate(PHASES):
         print(f"  {i} {name:<15} {buf[i]:>16}  {100.0 * buf[i] / max(total, 1):6.2f}%")
+    print(f"  lookback walk iterations={buf[6]}  publish stalls={buf[7]}")
 
 
 if __name__ == "__main__":

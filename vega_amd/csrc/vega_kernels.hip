@@ -634,7 +634,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
              * of all wave cycles). The slot is one relaxed agent-scope
              * 8-byte atomicMax granule — monotone because inclusive counts
              * are nondecreasing in tile id. */
-            gdesc_t *slot = (gdesc_t *)(ff + t);
+            gdesc_t *slot = (gdesc_t *)(ff + (size_t)t * 8); /* own 64B line */
             unsigned long long snap =
                 __hip_atomic_load(slot, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
             int64_t m = (int64_t)(snap >> 42);
@@ -642,7 +642,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
             if (m > (int64_t)vb) { m = 0; base = 0; } /* snapshot from a later tile: unusable */
             bool hit_inc = false;
             int64_t j = (int64_t)vb - 1;
-            uint32_t spins = 0;
+            uint32_t spins = 0, niter = 0, nstall = 0;
             while (j >= m) {
                 /* bounded spin: a lost predecessor can never wedge the GPU —
                  * set the abort flag, bail, and let the host fail loudly */
@@ -669,13 +669,21 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                     break;
                 }
                 if (done) break;
-                if (stall) __builtin_amdgcn_s_sleep(1);
+                if (stall) { nstall++; __builtin_amdgcn_s_sleep(1); }
+                niter++;
             }
             if (!hit_inc) excl_tiles += base;
+            if (phc) { /* diagnostic: walk iterations vs publish stalls */
+                atomicAdd(&phc[6], (unsigned long long)niter);
+                atomicAdd(&phc[7], (unsigned long long)nstall);
+            }
         }
-        __hip_atomic_fetch_max((gdesc_t *)(ff + t),
-                               ((unsigned long long)(vb + 1) << 42) | (excl_tiles + cnt),
-                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        /* sparse publish (every 8th tile) on padded lines: bounds the
+         * snapshot lag at 8 tiles while dividing slot contention 8x */
+        if ((vb & 7) == 7 || vb + 1 == nblocks)
+            __hip_atomic_fetch_max((gdesc_t *)(ff + (size_t)t * 8),
+                                   ((unsigned long long)(vb + 1) << 42) | (excl_tiles + cnt),
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            (excl_tiles + cnt) | OSW_ST_INC,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -755,7 +763,7 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     unsigned long long *phc = phase_prof_buf();
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
-    HIP_TRY(hipMemsetAsync(ff, 0, 256 * 8, s));
+    HIP_TRY(hipMemsetAsync(ff, 0, 256 * 64, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
@@ -856,7 +864,7 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_abort = (int *)ws.take(256);
-    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 8);
+    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 64);
     if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort || !ff_d)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
@@ -1091,7 +1099,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
     int *d_abort = (int *)ws.take(256);
-    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 8);
+    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 64);
     uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
     unsigned long long *wl = (unsigned long long *)ws.take(CLEANUP_WL_CAP * 8);
     uint32_t *wl_count = (uint32_t *)ws.take(256);
