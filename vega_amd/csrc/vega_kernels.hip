@@ -647,15 +647,24 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 /* bounded spin: a lost predecessor can never wedge the GPU —
                  * set the abort flag, bail, and let the host fail loudly */
                 if (++spins > (1u << 26)) { *d_abort = 1; break; }
-                /* probe up to 4 predecessors with independent loads */
-                unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
-                int navail = (j - m >= 3) ? 4 : (int)(j - m + 1);
+                /* probe up to 8 predecessors with independent loads in
+                 * flight: the walk front tracks the publish front at
+                 * (probe depth) tiles per L2 round trip, so depth divides
+                 * the iteration count (diagnostics: 58 iterations ≈ 230-tile
+                 * window / 4 at depth 4). Explicit scalars — a 16-wide
+                 * indexed array spilled to scratch and regressed 40%. */
+                unsigned long long d0, d1 = 0, d2 = 0, d3 = 0, d4 = 0, d5 = 0, d6 = 0, d7 = 0;
+                int navail = (j - m >= 7) ? 8 : (int)(j - m + 1);
                 d0 = __hip_atomic_load(col + j * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 if (navail > 1) d1 = __hip_atomic_load(col + (j - 1) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 if (navail > 2) d2 = __hip_atomic_load(col + (j - 2) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 if (navail > 3) d3 = __hip_atomic_load(col + (j - 3) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 4) d4 = __hip_atomic_load(col + (j - 4) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 5) d5 = __hip_atomic_load(col + (j - 5) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 6) d6 = __hip_atomic_load(col + (j - 6) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (navail > 7) d7 = __hip_atomic_load(col + (j - 7) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 bool done = false, stall = false;
-                unsigned long long dd4[4] = {d0, d1, d2, d3};
+                unsigned long long dd4[8] = {d0, d1, d2, d3, d4, d5, d6, d7};
                 for (int q = 0; q < navail; ++q) {
                     unsigned long long st = dd4[q] >> 62;
                     if (st == 2) { /* INC covers tiles [0, j]: base unused */
