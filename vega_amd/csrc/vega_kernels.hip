@@ -605,6 +605,15 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            (unsigned long long)cnt | OSW_ST_AGG,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        /* second level: add this tile's count into its 32-tile GROUP slot
+         * (arrivals in the high bits, sum in the low) — one relaxed 8-byte
+         * atomicAdd granule. A full group (arrivals == 32) is consumable
+         * with ONE load, and group sums need only the RANK front — they
+         * never wait on predecessors' lookbacks, which is what breaks the
+         * walk-length/retire-front feedback the flat walk suffered from. */
+        __hip_atomic_fetch_add((gdesc_t *)&ff[((uint64_t)vb >> 5) * 256 + t],
+                               (1ULL << 42) | (unsigned long long)cnt,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
     uint32_t inc = cnt;
     for (int off = 1; off < 64; off <<= 1) {
@@ -627,52 +636,27 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         unsigned long long excl_tiles = 0;
         if (vb > 0) {
             gdesc_t *col = (gdesc_t *)(desc + t);
-            /* FAST-FORWARD: ff[t] holds (m << 42 | count) = the largest
-             * published inclusive prefix for this digit — count covers tiles
-             * [0, m). Snapshot it and walk only (m .. vb-1], not all ~512
-             * resident predecessors (phase profiling: the full walk was 62%
-             * of all wave cycles). The slot is one relaxed agent-scope
-             * 8-byte atomicMax granule — monotone because inclusive counts
-             * are nondecreasing in tile id. */
-            gdesc_t *slot = (gdesc_t *)(ff + (size_t)t * 8); /* own 64B line */
-            unsigned long long snap =
-                __hip_atomic_load(slot, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-            int64_t m = (int64_t)(snap >> 42);
-            unsigned long long base = snap & ((1ULL << 42) - 1);
-            if (m > (int64_t)vb) { m = 0; base = 0; } /* snapshot from a later tile: unusable */
-            bool hit_inc = false;
-            int64_t j = (int64_t)vb - 1;
+            gdesc_t *gcol = (gdesc_t *)(ff + t);
             uint32_t spins = 0, niter = 0, nstall = 0;
-            while (j >= m) {
+            bool done = false;
+            /* 1) singles within the own 32-tile group (< 32 of them) */
+            int64_t gb_lo = (int64_t)(vb & ~31u);
+            int64_t j = (int64_t)vb - 1;
+            while (j >= gb_lo) {
                 /* bounded spin: a lost predecessor can never wedge the GPU —
                  * set the abort flag, bail, and let the host fail loudly */
-                if (++spins > (1u << 26)) { *d_abort = 1; break; }
-                /* probe up to 8 predecessors with independent loads in
-                 * flight: the walk front tracks the publish front at
-                 * (probe depth) tiles per L2 round trip, so depth divides
-                 * the iteration count (diagnostics: 58 iterations ≈ 230-tile
-                 * window / 4 at depth 4). Explicit scalars — a 16-wide
-                 * indexed array spilled to scratch and regressed 40%. */
-                unsigned long long d0, d1 = 0, d2 = 0, d3 = 0, d4 = 0, d5 = 0, d6 = 0, d7 = 0;
-                int navail = (j - m >= 7) ? 8 : (int)(j - m + 1);
+                if (++spins > (1u << 26)) { *d_abort = 1; done = true; break; }
+                unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
+                int navail = (j - gb_lo >= 3) ? 4 : (int)(j - gb_lo + 1);
                 d0 = __hip_atomic_load(col + j * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 if (navail > 1) d1 = __hip_atomic_load(col + (j - 1) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 if (navail > 2) d2 = __hip_atomic_load(col + (j - 2) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 if (navail > 3) d3 = __hip_atomic_load(col + (j - 3) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 4) d4 = __hip_atomic_load(col + (j - 4) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 5) d5 = __hip_atomic_load(col + (j - 5) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 6) d6 = __hip_atomic_load(col + (j - 6) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                if (navail > 7) d7 = __hip_atomic_load(col + (j - 7) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                bool done = false, stall = false;
-                unsigned long long dd4[8] = {d0, d1, d2, d3, d4, d5, d6, d7};
+                bool stall = false;
+                unsigned long long dd4[4] = {d0, d1, d2, d3};
                 for (int q = 0; q < navail; ++q) {
                     unsigned long long st = dd4[q] >> 62;
-                    if (st == 2) { /* INC covers tiles [0, j]: base unused */
-                        excl_tiles += dd4[q] & OSW_CNT_MASK;
-                        hit_inc = true;
-                        done = true;
-                        break;
-                    }
+                    if (st == 2) { excl_tiles += dd4[q] & OSW_CNT_MASK; done = true; break; }
                     if (st == 1) { excl_tiles += dd4[q] & OSW_CNT_MASK; j--; continue; }
                     stall = true;
                     break;
@@ -681,18 +665,38 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 if (stall) { nstall++; __builtin_amdgcn_s_sleep(1); }
                 niter++;
             }
-            if (!hit_inc) excl_tiles += base;
+            /* 2) whole groups below: one load consumes 32 ranked tiles;
+             * the group's last-tile descriptor doubles as the deep-history
+             * INC shortcut */
+            if (!done && vb >= 32) {
+                int64_t g = (int64_t)(vb >> 5) - 1;
+                while (g >= 0) {
+                    if (++spins > (1u << 26)) { *d_abort = 1; break; }
+                    unsigned long long ge = __hip_atomic_load(
+                        col + ((uint64_t)g * 32 + 31) * 256, __ATOMIC_RELAXED,
+                        __HIP_MEMORY_SCOPE_AGENT);
+                    unsigned long long gd = __hip_atomic_load(
+                        gcol + (uint64_t)g * 256, __ATOMIC_RELAXED,
+                        __HIP_MEMORY_SCOPE_AGENT);
+                    if ((ge >> 62) == 2) { /* INC: covers [0, g*32+31] */
+                        excl_tiles += ge & OSW_CNT_MASK;
+                        break;
+                    }
+                    if ((gd >> 42) == 32) { /* all 32 tiles ranked: final sum */
+                        excl_tiles += gd & ((1ULL << 42) - 1);
+                        g--;
+                        continue;
+                    }
+                    nstall++;
+                    __builtin_amdgcn_s_sleep(1);
+                    niter++;
+                }
+            }
             if (phc) { /* diagnostic: walk iterations vs publish stalls */
                 atomicAdd(&phc[6], (unsigned long long)niter);
                 atomicAdd(&phc[7], (unsigned long long)nstall);
             }
         }
-        /* sparse publish (every 8th tile) on padded lines: bounds the
-         * snapshot lag at 8 tiles while dividing slot contention 8x */
-        if ((vb & 7) == 7 || vb + 1 == nblocks)
-            __hip_atomic_fetch_max((gdesc_t *)(ff + (size_t)t * 8),
-                                   ((unsigned long long)(vb + 1) << 42) | (excl_tiles + cnt),
-                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            (excl_tiles + cnt) | OSW_ST_INC,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -772,7 +776,7 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     unsigned long long *phc = phase_prof_buf();
     uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
-    HIP_TRY(hipMemsetAsync(ff, 0, 256 * 64, s));
+    HIP_TRY(hipMemsetAsync(ff, 0, (size_t)((nb + 31) / 32) * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
@@ -873,7 +877,8 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_abort = (int *)ws.take(256);
-    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 64);
+    unsigned long long *ff_d =
+        (unsigned long long *)ws.take((size_t)((nb + 31) / 32) * 256 * 8);
     if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort || !ff_d)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
@@ -1108,7 +1113,8 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
     int *d_abort = (int *)ws.take(256);
-    unsigned long long *ff_d = (unsigned long long *)ws.take(256 * 64);
+    unsigned long long *ff_d =
+        (unsigned long long *)ws.take((size_t)((nb + 31) / 32) * 256 * 8);
     uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
     unsigned long long *wl = (unsigned long long *)ws.take(CLEANUP_WL_CAP * 8);
     uint32_t *wl_count = (uint32_t *)ws.take(256);
@@ -1800,6 +1806,7 @@ size_t ws_bytes_for(uint64_t n) {
     /* scan recursion partials: nb/TILE + nb/TILE^2 + ... < nb/2048 */
     b += (((size_t)nb / 2048 + 4096) * 4 + 255) & ~255ULL;
     b += ((size_t)257 * 4 + 255) & ~255ULL;       /* partition starts */
+    b += ((size_t)((nb + 31) / 32) * 2048 + 255) & ~255ULL; /* group descriptors */
     b += (size_t)CLEANUP_WL_CAP * 8 + 512;        /* cleanup long-run worklist */
     b += ((size_t)nb * BLOCK * 12 + 255) & ~255ULL; /* f64 lead partials (seg, OP 2) */
     b += 1 << 20;                                 /* slack */
