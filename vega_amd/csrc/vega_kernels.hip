@@ -995,35 +995,17 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
                                 unsigned long long *wl, uint32_t *wl_count) {
     uint64_t nchunks = (n + 3) / 4;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    /* 4 strided chunks per iteration: the h32 scan is latency-bound at one
-     * 16-B load in flight per lane (measured 600 GB/s); batching the loads
-     * up front quadruples outstanding memory per lane */
-    for (uint64_t cb = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; cb < nchunks;
-         cb += 4 * stride) {
-        uint4 hv4[4];
-        uint32_t hp4[4];
-        int mm4[4];
-#pragma unroll
-        for (int u = 0; u < 4; ++u) {
-            uint64_t c = cb + (uint64_t)u * stride;
-            if (c >= nchunks) { mm4[u] = 0; continue; }
-            uint64_t i0 = 4 * c;
-            mm4[u] = (int)((n - i0 < 4) ? (n - i0) : 4);
-            if (mm4[u] == 4) hv4[u] = ((const uint4 *)h32)[c];
-            hp4[u] = (i0 > 0) ? h32[i0 - 1] : 0;
-        }
-        for (int u = 0; u < 4; ++u) {
-        uint64_t c = cb + (uint64_t)u * stride;
-        int m = mm4[u];
-        if (m == 0) continue;
+    for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
         uint64_t i0 = 4 * c;
+        int m = (int)((n - i0 < 4) ? (n - i0) : 4);
         uint32_t hh[5];
         if (m == 4) {
-            hh[1] = hv4[u].x; hh[2] = hv4[u].y; hh[3] = hv4[u].z; hh[4] = hv4[u].w;
+            uint4 hv = ((const uint4 *)h32)[c];
+            hh[1] = hv.x; hh[2] = hv.y; hh[3] = hv.z; hh[4] = hv.w;
         } else {
             for (int j = 0; j < m; ++j) hh[j + 1] = h32[i0 + j];
         }
-        hh[0] = (i0 > 0) ? hp4[u] : ~hh[1]; /* sentinel differs */
+        hh[0] = (i0 > 0) ? h32[i0 - 1] : ~hh[1]; /* sentinel differs */
         for (int j = 0; j < m; ++j) {
             uint64_t gi = i0 + j;
             if (gi != 0 && hh[j + 1] == hh[j]) continue; /* not a run start */
@@ -1072,7 +1054,6 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
                 v[y] = vx;
             }
         }
-        } /* u */
     }
 }
 
