@@ -487,17 +487,6 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
 
 #define OSW_GRP_LG 4
 #define OSW_GRP (1 << OSW_GRP_LG)
-/* onesweep scatter geometry: 1024 threads x 8 rows = 8192-row tiles (one
- * workgroup per CU, 16 waves — same occupancy as 2x512, half the tiles, so
- * half the lookback instances and descriptor traffic; writeout runs per
- * digit double, improving store coalescing) */
-#define OSW_SB 1024
-#define OSW_SW (OSW_SB / 64)
-#define OSW_SIPT 8
-#define OSW_TILE (OSW_SB * OSW_SIPT)
-static inline uint32_t osw_nblocks_for(uint64_t n) {
-    return (uint32_t)((n + OSW_TILE - 1) / OSW_TILE);
-}
 #define OSW_ST_AGG (1ULL << 62)
 #define OSW_ST_INC (2ULL << 62)
 #define OSW_CNT_MASK ((1ULL << 62) - 1)
@@ -505,7 +494,7 @@ static inline uint32_t osw_nblocks_for(uint64_t n) {
 typedef __attribute__((address_space(1))) unsigned long long gdesc_t;
 
 template <class DF, bool HAS_VALS, bool IN_PK, bool OUT_PK>
-__global__ __launch_bounds__(OSW_SB) void k_scatter_osw(
+__global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
     uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, int *d_abort,
@@ -517,15 +506,14 @@ __global__ __launch_bounds__(OSW_SB) void k_scatter_osw(
     /* 512 threads = 8 waves per block (16 waves/CU at 2 blocks): each wave
      * ranks a 512-row chunk of the 4096-row tile. IN_PK/OUT_PK: interleaved
      * (k,v) rows — one 16-B vector access per row. */
-    constexpr int SB = OSW_SB, SW = OSW_SW, SIPT = OSW_SIPT;
-    constexpr int OTILE = OSW_TILE;
+    constexpr int SB = 512, SW = 8, SIPT = 8;
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    ulonglong2 *spk = (ulonglong2 *)smem;                       /* OTILE 16B */
+    ulonglong2 *spk = (ulonglong2 *)smem;                       /* TILE 16B */
     uint64_t *sk = (uint64_t *)smem;                            /* no-vals */
-    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 16 : 8) * (size_t)OTILE);
+    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 16 : 8) * (size_t)TILE);
     uint32_t *whist = hist + 256;                               /* SW*256 */
-    uint32_t *wsc = whist + SW * 256;                           /* SW */
-    uint32_t *vbp = wsc + SW;                                   /* 8 */
+    uint32_t *wsc = whist + SW * 256;                           /* 8 */
+    uint32_t *vbp = wsc + 8;                                    /* 8 */
     uint32_t *tilebase = vbp + 8;                               /* 256 */
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
@@ -542,8 +530,8 @@ __global__ __launch_bounds__(OSW_SB) void k_scatter_osw(
     for (int i = t; i < SW * 256; i += SB) whist[i] = 0;
     __syncthreads();
     const uint32_t vb = vbp[0];
-    const uint64_t tbase = (uint64_t)vb * OTILE;
-    const uint32_t tile_n = (uint32_t)((n - tbase < OTILE) ? (n - tbase) : OTILE);
+    const uint64_t tbase = (uint64_t)vb * TILE;
+    const uint32_t tile_n = (uint32_t)((n - tbase < TILE) ? (n - tbase) : TILE);
 
     /* prefetch the wave's whole 512-row chunk (independent loads in flight
      * together: ONE memory latency per chunk, not one per round) */
@@ -799,37 +787,26 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     unsigned long long *phc = phase_prof_buf();
-    uint32_t nb = osw_nblocks_for(n);
+    uint32_t nb = nblocks_for(n);
     HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ff, 0, (size_t)((nb + OSW_GRP - 1) / OSW_GRP) * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
-    size_t sh = (has_vals ? 16 : 8) * (size_t)OSW_TILE +
-                (256 + OSW_SW * 256 + OSW_SW + 8 + 256) * 4;
+    size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
-        (void)hipFuncSetAttribute((const void *)&k_scatter_osw<DF, false, false, false>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(OSW_SB), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
                            in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
     } else if (!in_pk && !out_pk) {
-        (void)hipFuncSetAttribute((const void *)&k_scatter_osw<DF, true, false, false>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(OSW_SB), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
                            in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, df);
     } else if (!in_pk && out_pk) {
-        (void)hipFuncSetAttribute((const void *)&k_scatter_osw<DF, true, false, true>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(OSW_SB), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
                            in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
     } else if (in_pk && out_pk) {
-        (void)hipFuncSetAttribute((const void *)&k_scatter_osw<DF, true, true, true>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(OSW_SB), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
                            in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
     } else {
-        (void)hipFuncSetAttribute((const void *)&k_scatter_osw<DF, true, true, false>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(OSW_SB), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
                            in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, df);
     }
     return hipGetLastError();
