@@ -314,13 +314,15 @@ __global__ void k_sample_keys(const uint64_t *k, uint64_t n, uint32_t ns,
     if (i < ns) out[i] = k[(uint64_t)i * (n / ns)];
 }
 
-/* all 8 byte-position histograms in one pass (radix pass skipping / the
- * onesweep global digit bases). HASHSRC hists the splitmix64 of the key.
- * Per-wave-private LDS copies (32 KB) cut hot-digit serialization 4x. */
-template <bool HASHSRC>
+/* NBYTES byte-position histograms in one pass (radix pass skipping / the
+ * onesweep global digit bases). HASHSRC hists the splitmix64 of the key —
+ * the hash-grouping sort only consumes its 4-5 low bytes, so NBYTES trims
+ * the dead atomics. Per-wave-private LDS copies cut hot-digit
+ * serialization 4x. */
+template <bool HASHSRC, int NBYTES>
 __global__ void k_hist8t(const uint64_t *keys, uint64_t n, uint32_t *h8) {
-    __shared__ uint32_t h[4][8][256];
-    for (int i = threadIdx.x; i < 4 * 8 * 256; i += BLOCK) ((uint32_t *)h)[i] = 0;
+    __shared__ uint32_t h[4][NBYTES][256];
+    for (int i = threadIdx.x; i < 4 * NBYTES * 256; i += BLOCK) ((uint32_t *)h)[i] = 0;
     __syncthreads();
     const int w = threadIdx.x >> 6;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
@@ -339,7 +341,7 @@ __global__ void k_hist8t(const uint64_t *keys, uint64_t n, uint32_t *h8) {
             int lane = threadIdx.x & 63;
             int nact = __popcll(act);
 #pragma unroll
-            for (int b = 0; b < 8; ++b) {
+            for (int b = 0; b < NBYTES; ++b) {
                 uint32_t d = (uint32_t)(k >> (8 * b)) & 0xFF;
                 uint32_t d0 = (uint32_t)__shfl((int)d, leader);
                 uint64_t same = __ballot(d == d0);
@@ -351,13 +353,14 @@ __global__ void k_hist8t(const uint64_t *keys, uint64_t n, uint32_t *h8) {
             }
         } else { /* hash bytes are uniform-random: the check never pays */
 #pragma unroll
-            for (int b = 0; b < 8; ++b) atomicAdd(&h[w][b][(k >> (8 * b)) & 0xFF], 1u);
+            for (int b = 0; b < NBYTES; ++b) atomicAdd(&h[w][b][(k >> (8 * b)) & 0xFF], 1u);
         }
     }
     __syncthreads();
     const uint32_t *hf = (const uint32_t *)h;
-    for (int i = threadIdx.x; i < 8 * 256; i += BLOCK) {
-        uint32_t v = hf[i] + hf[2048 + i] + hf[4096 + i] + hf[6144 + i];
+    constexpr int NW = NBYTES * 256;
+    for (int i = threadIdx.x; i < NW; i += BLOCK) {
+        uint32_t v = hf[i] + hf[NW + i] + hf[2 * NW + i] + hf[3 * NW + i];
         if (v) atomicAdd(&h8[i], v);
     }
 }
@@ -489,7 +492,8 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
 #define OSW_GRP (1 << OSW_GRP_LG)
 #define OSW_ST_AGG (1ULL << 62)
 #define OSW_ST_INC (2ULL << 62)
-#define OSW_CNT_MASK ((1ULL << 62) - 1)
+#define OSW_CNT_MASK ((1ULL << 56) - 1)
+#define OSW_TAG(d) ((uint32_t)((d) >> 56) & 0x3F)
 
 typedef __attribute__((address_space(1))) unsigned long long gdesc_t;
 
@@ -498,7 +502,10 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
     const uint64_t *in_k, const uint64_t *in_v, uint64_t n, uint32_t nblocks,
     const uint32_t *gbase, unsigned long long *desc, uint32_t *ticket,
     uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out, int *d_abort,
-    unsigned long long *ff, unsigned long long *phc, DF df) {
+    unsigned long long *ff, unsigned long long *phc, int ptag, DF df) {
+    /* ptag: this pass's 6-bit descriptor tag — descriptors are zeroed ONCE
+     * per sort call, and a word is valid only when its tag matches, so the
+     * 0.5 GB per-pass desc memset disappears. */
     /* phc (diagnostic builds, VEGA_PHASE_PROF=1): per-phase shader-cycle
      * sums, one sample per wave — phases: 0 prefetch, 1 rank, 2 publish+
      * starts, 3 lookback, 4 reorder, 5 writeout (s_memtime; the microarch
@@ -605,7 +612,8 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
          * 2 KB burst (digit-major measured 3.6x slower — 256 scattered
          * lines per publish) */
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
-                           (unsigned long long)cnt | OSW_ST_AGG,
+                           (unsigned long long)cnt | OSW_ST_AGG |
+                               ((unsigned long long)ptag << 56),
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         /* second level: add this tile's count into its 32-tile GROUP slot
          * (arrivals in the high bits, sum in the low) — one relaxed 8-byte
@@ -657,7 +665,8 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 bool stall = false;
                 unsigned long long dd4[4] = {d0, d1, d2, d3};
                 for (int q = 0; q < navail; ++q) {
-                    unsigned long long st = dd4[q] >> 62;
+                    unsigned long long st =
+                        (OSW_TAG(dd4[q]) == (uint32_t)ptag) ? (dd4[q] >> 62) : 0;
                     if (st == 2) { excl_tiles += dd4[q] & OSW_CNT_MASK; done = true; break; }
                     if (st == 1) { excl_tiles += dd4[q] & OSW_CNT_MASK; j--; continue; }
                     stall = true;
@@ -686,7 +695,8 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                     unsigned long long ge = __hip_atomic_load(
                         col + (((uint64_t)g + 1) * OSW_GRP - 1) * 256, __ATOMIC_RELAXED,
                         __HIP_MEMORY_SCOPE_AGENT);
-                    if ((ge >> 62) == 2) { /* INC: covers [0, (g+1)*GRP-1] */
+                    if ((ge >> 62) == 2 && OSW_TAG(ge) == (uint32_t)ptag) {
+                        /* INC: covers [0, (g+1)*GRP-1] */
                         excl_tiles += ge & OSW_CNT_MASK;
                         break;
                     }
@@ -711,7 +721,8 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
             }
         }
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
-                           (excl_tiles + cnt) | OSW_ST_INC,
+                           ((excl_tiles + cnt) | OSW_ST_INC) |
+                               ((unsigned long long)ptag << 56),
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
     }
@@ -748,14 +759,19 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         }
         uint32_t d = df(k);
         uint64_t gpos = (uint64_t)tilebase[d] + (p - hist[d]);
+        /* nontemporal: scattered digit-region stores are never re-read this
+         * pass — keep them out of L2 */
         if (OUT_PK) {
-            ((ulonglong2 *)out_k)[gpos] = make_ulonglong2(k, v);
+            typedef unsigned long long vull2 __attribute__((ext_vector_type(2)));
+            vull2 kv2 = {k, v};
+            __builtin_nontemporal_store(kv2, (vull2 *)&((ulonglong2 *)out_k)[gpos]);
         } else {
-            out_k[gpos] = k;
-            if (HAS_VALS) out_v[gpos] = v;
+            __builtin_nontemporal_store(k, &out_k[gpos]);
+            if (HAS_VALS) __builtin_nontemporal_store(v, &out_v[gpos]);
             if (h32_out) /* low hash bits for the grouping cleanup (4 B/row
                             beats its re-hashing 16 B/row) */
-                h32_out[gpos] = (uint32_t)vega_hash_u64(k);
+                __builtin_nontemporal_store((uint32_t)vega_hash_u64(k),
+                                            &h32_out[gpos]);
         }
     }
     VEGA_PHASE_MARK(5)
@@ -783,31 +799,31 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    uint64_t n, const uint32_t *gbase_d,
                                    unsigned long long *desc, uint32_t *ticket,
                                    uint64_t *out_k, uint64_t *out_v, uint32_t *h32_out,
-                                   int *d_abort, unsigned long long *ff,
+                                   int *d_abort, unsigned long long *ff, int ptag,
                                    bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     unsigned long long *phc = phase_prof_buf();
     uint32_t nb = nblocks_for(n);
-    HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
+    /* desc is zeroed once per SORT call (pass tags invalidate stale words) */
     HIP_TRY(hipMemsetAsync(ff, 0, (size_t)((nb + OSW_GRP - 1) / OSW_GRP) * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
     if (!has_vals) {
         hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, ptag, df);
     } else if (!in_pk && !out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, ptag, df);
     } else if (!in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
+                           in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, ptag, df);
     } else if (in_pk && out_pk) {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, ptag, df);
     } else {
         hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
-                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, df);
+                           in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, ptag, df);
     }
     return hipGetLastError();
 }
@@ -895,13 +911,15 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort || !ff_d)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
+    HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
+    int ptag_ctr = 0;
 
     /* exact per-byte histograms: pass skipping + the onesweep global bases */
     HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
     {
         ProfScope ps("hist8", s);
         uint32_t gb = nb < 2048 ? nb : 2048;
-        hipLaunchKernelGGL(k_hist8t<false>, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+        hipLaunchKernelGGL((k_hist8t<false, 8>), dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
         HIP_TRY(hipGetLastError());
     }
     uint32_t hh[8 * 256];
@@ -940,11 +958,11 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         if (p == 7 && signed_order) {
             RadixDigitTopSigned df{56};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, d_abort, ff_d, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, ff_d, ptag_ctr++, has_vals, in_pk, out_pk, df, "radix_scatter"));
         } else {
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, src_k, src_v, n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, d_abort, ff_d, has_vals, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, ff_d, ptag_ctr++, has_vals, in_pk, out_pk, df, "radix_scatter"));
         }
         cur = dk;
     }
@@ -1135,6 +1153,8 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         !ff_d || !h32buf || !wl || !wl_count)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
+    HIP_TRY(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
+    int ptag_ctr = 0;
     /* strict order (full (h32,key) lex within runs) only when a caller will
      * binary-search the result (joins/cogroup: force_hbytes or order_tag);
      * the reduce path needs only equal-keys-adjacent (relaxed) */
@@ -1146,15 +1166,17 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 
     /* exact byte histograms of SRC(k); fills hh and gb_host (functor order
      * == raw order here), returns #active passes */
-    auto exact_hists = [&](bool hashsrc, int *active) -> hipError_t {
+    auto exact_hists = [&](bool hashsrc, int nbl, int *active) -> hipError_t {
         HIP_TRY(hipMemsetAsync(h8, 0, 8 * 256 * 4, s));
         {
             ProfScope ps(hashsrc ? "ghist" : "hist8", s);
             uint32_t gb = nb < 2048 ? nb : 2048;
-            if (hashsrc)
-                hipLaunchKernelGGL(k_hist8t<true>, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+            if (hashsrc && nbl == 4)
+                hipLaunchKernelGGL((k_hist8t<true, 4>), dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+            else if (hashsrc)
+                hipLaunchKernelGGL((k_hist8t<true, 5>), dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
             else
-                hipLaunchKernelGGL(k_hist8t<false>, dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
+                hipLaunchKernelGGL((k_hist8t<false, 8>), dim3(gb), dim3(BLOCK), 0, s, in_k, n, h8);
             HIP_TRY(hipGetLastError());
         }
         HIP_TRY(hipMemcpyAsync(hh, h8, 8 * 256 * 4, hipMemcpyDeviceToHost, s));
@@ -1190,7 +1212,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
             RadixDigit df{8 * p};
             HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur, i == 0 ? in_v : nullptr,
                                      n, gbase_d + p * 256, desc, ticket,
-                                     dk, dv, nullptr, d_abort, ff_d, true, in_pk, out_pk, df, "radix_scatter"));
+                                     dk, dv, nullptr, d_abort, ff_d, ptag_ctr++, true, in_pk, out_pk, df, "radix_scatter"));
             cur = dk;
         }
         *rk = cur;
@@ -1205,7 +1227,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     const uint64_t *cur_k = in_k, *cur_v = in_v;
     if (a_est <= 5) {
         int active = 8;
-        HIP_TRY(exact_hists(false, &active));
+        HIP_TRY(exact_hists(false, 8, &active));
         if (active <= 5) { /* narrow keys: skipped key sort groups exactly */
             HIP_TRY(run_key_passes(res_k, res_v));
             int ab = 0;
@@ -1222,7 +1244,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
      * it the order contract) for joins. */
     const int hbytes = force_hbytes ? force_hbytes : ((n <= (1ULL << 30)) ? 4 : 5);
     int active5 = 0;
-    HIP_TRY(exact_hists(true, &active5));
+    HIP_TRY(exact_hists(true, hbytes, &active5));
     for (int i = 0; i < hbytes; ++i) {
         bool in_pk = i > 0, out_pk = i < hbytes - 1;
         uint64_t *dbuf = ((i & 1) == 0) ? pA : pB;
@@ -1232,7 +1254,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur_k, i == 0 ? in_v : nullptr,
                                  n, gbase_d + i * 256, desc, ticket,
                                  dk, dv, out_pk ? nullptr : h32buf,
-                                 d_abort, ff_d, true, in_pk, out_pk, df, "radix_scatter"));
+                                 d_abort, ff_d, ptag_ctr++, true, in_pk, out_pk, df, "radix_scatter"));
         cur_k = dk;
         cur_v = out_pk ? nullptr : dk + n;
     }
@@ -1254,7 +1276,7 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     HIP_TRY(hipStreamSynchronize(s));
     if (err) { /* an oversized hash-dirty run: full key sort instead */
         int active = 8;
-        HIP_TRY(exact_hists(false, &active));
+        HIP_TRY(exact_hists(false, 8, &active));
         HIP_TRY(run_key_passes(&cur_k, &cur_v));
     }
     int ab = 0;
