@@ -759,19 +759,17 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         }
         uint32_t d = df(k);
         uint64_t gpos = (uint64_t)tilebase[d] + (p - hist[d]);
-        /* nontemporal: scattered digit-region stores are never re-read this
-         * pass — keep them out of L2 */
+        /* plain stores: L2 write-combining of the 16-B digit-run stores is
+         * load-bearing — nontemporal stores regressed the pass 33% (partial
+         * 64-B sectors straight to DRAM) */
         if (OUT_PK) {
-            typedef unsigned long long vull2 __attribute__((ext_vector_type(2)));
-            vull2 kv2 = {k, v};
-            __builtin_nontemporal_store(kv2, (vull2 *)&((ulonglong2 *)out_k)[gpos]);
+            ((ulonglong2 *)out_k)[gpos] = make_ulonglong2(k, v);
         } else {
-            __builtin_nontemporal_store(k, &out_k[gpos]);
-            if (HAS_VALS) __builtin_nontemporal_store(v, &out_v[gpos]);
+            out_k[gpos] = k;
+            if (HAS_VALS) out_v[gpos] = v;
             if (h32_out) /* low hash bits for the grouping cleanup (4 B/row
                             beats its re-hashing 16 B/row) */
-                __builtin_nontemporal_store((uint32_t)vega_hash_u64(k),
-                                            &h32_out[gpos]);
+                h32_out[gpos] = (uint32_t)vega_hash_u64(k);
         }
     }
     VEGA_PHASE_MARK(5)
