@@ -1014,7 +1014,7 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
     for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
         uint64_t i0 = 4 * c;
         int m = (int)((n - i0 < 4) ? (n - i0) : 4);
-        uint32_t hh[5];
+        uint32_t hh[6];
         if (m == 4) {
             uint4 hv = ((const uint4 *)h32)[c];
             hh[1] = hv.x; hh[2] = hv.y; hh[3] = hv.z; hh[4] = hv.w;
@@ -1022,11 +1022,15 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
             for (int j = 0; j < m; ++j) hh[j + 1] = h32[i0 + j];
         }
         hh[0] = (i0 > 0) ? h32[i0 - 1] : ~hh[1]; /* sentinel differs */
+        /* peek one past the window too: chunk-tail single-row runs then
+         * resolve without a dependent walk load (issued together with the
+         * uint4, not after it) */
+        hh[m + 1] = (i0 + m < n) ? h32[i0 + m] : ~hh[m];
         for (int j = 0; j < m; ++j) {
             uint64_t gi = i0 + j;
             if (gi != 0 && hh[j + 1] == hh[j]) continue; /* not a run start */
-            /* peek in-window: single-row runs need no global traffic */
-            if (j + 1 < m && hh[j + 2] != hh[j + 1]) continue;
+            /* peek: single-row runs need no further global traffic */
+            if (hh[j + 2] != hh[j + 1]) continue;
             uint64_t je = gi + 1;
             uint64_t wcap = gi + CLEANUP_WALK_CAP;
             while (je < n && je < wcap && h32[je] == hh[j + 1]) je++;
@@ -1269,18 +1273,23 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
                            cur_k, h32buf, n, strict, d_err, wl, wl_count);
         HIP_TRY(hipGetLastError());
     }
-    int err = 0;
-    HIP_TRY(hipMemcpyAsync(&err, d_err, 4, hipMemcpyDeviceToHost, s));
+    /* one readback for both flags: abort is final once the scatter passes
+     * and cleanup ahead of this sync have been drained */
+    int flags[2] = {0, 0};
+    HIP_TRY(hipMemcpyAsync(&flags[0], d_err, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipMemcpyAsync(&flags[1], d_abort, 4, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
+    if (flags[1]) return hipErrorUnknown; /* lookback bailed: fail loudly */
+    int err = flags[0];
     if (err) { /* an oversized hash-dirty run: full key sort instead */
         int active = 8;
         HIP_TRY(exact_hists(false, 8, &active));
         HIP_TRY(run_key_passes(&cur_k, &cur_v));
+        int ab2 = 0;
+        HIP_TRY(hipMemcpyAsync(&ab2, d_abort, 4, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        if (ab2) return hipErrorUnknown;
     }
-    int ab = 0;
-    HIP_TRY(hipMemcpyAsync(&ab, d_abort, 4, hipMemcpyDeviceToHost, s));
-    HIP_TRY(hipStreamSynchronize(s));
-    if (ab) return hipErrorUnknown; /* lookback bailed: fail loudly */
     if (order_tag && !err) *order_tag = (hbytes == 4) ? 4 : 0;
     *res_k = cur_k;
     *res_v = cur_v;
