@@ -385,12 +385,10 @@ static int reduce_multi(vega_ctx *c, RddImpl *r, int op, uint32_t nparts,
         if (rc) return rc;
         CTX_TRY(c, hipSetDevice(p));
         Ws ws(c->mws[p], c->mws_bytes[p]);
-        const uint64_t *sk, *sv;
-        CTX_TRY(c, group_sort_u64(c->mstreams[p], (const uint64_t *)rk[p],
-                                  (const uint64_t *)rv[p], rn[p], 0, nullptr, ws, &sk, &sv));
         uint64_t nout = 0;
-        CTX_TRY(c, seg_reduce(c->mstreams[p], sk, sv, rn[p], op,
-                              (uint64_t *)o->mk[p], o->mv[p], &nout, ws));
+        CTX_TRY(c, group_sort_reduce(c->mstreams[p], (const uint64_t *)rk[p],
+                                     (const uint64_t *)rv[p], rn[p], op,
+                                     (uint64_t *)o->mk[p], o->mv[p], &nout, ws));
         o->mn[p] = nout;
         total += nout;
     }
@@ -421,11 +419,10 @@ static int reduce_common(vega_ctx *c, vega_rdd_t rdd, int op, uint32_t nparts,
     rc = new_rdd(c, r->n ? r->n : 1, op == VEGA_OP_SUM_F64 ? 1 : 0, nparts, &o, out);
     if (rc) return rc;
     Ws ws(c->ws, c->ws_bytes);
-    const uint64_t *sk, *sv;
-    CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
-                              r->n, 0, nullptr, ws, &sk, &sv));
     uint64_t nout = 0;
-    CTX_TRY(c, seg_reduce(c->stream, sk, sv, r->n, op, (uint64_t *)o->d_k, o->d_v, &nout, ws));
+    CTX_TRY(c, group_sort_reduce(c->stream, (const uint64_t *)r->d_k,
+                                 (const uint64_t *)r->d_v, r->n, op,
+                                 (uint64_t *)o->d_k, o->d_v, &nout, ws));
     o->n = nout;
     o->sorted = false; /* grouped (hash order), not key-sorted; collect compares sorted */
     return VEGA_OK;
@@ -479,12 +476,10 @@ int vega_gpu_count_by_value(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts,
     rc = new_rdd(c, r->n ? r->n : 1, 0, nparts, &o, out);
     if (rc) return rc;
     Ws ws(c->ws, c->ws_bytes);
-    const uint64_t *sk, *sv;
-    CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_v, (const uint64_t *)r->d_k,
-                              r->n, 0, nullptr, ws, &sk, &sv));
     uint64_t nout = 0;
-    CTX_TRY(c, seg_reduce(c->stream, sk, sv, r->n, VEGA_OP_COUNT,
-                          (uint64_t *)o->d_k, o->d_v, &nout, ws));
+    CTX_TRY(c, group_sort_reduce(c->stream, (const uint64_t *)r->d_v,
+                                 (const uint64_t *)r->d_k, r->n, VEGA_OP_COUNT,
+                                 (uint64_t *)o->d_k, o->d_v, &nout, ws));
     o->n = nout;
     return VEGA_OK;
 }
@@ -1025,11 +1020,10 @@ int vega_dev_sort_reduce(void *stream, const int64_t *in_k, const void *in_v,
                          uint64_t n, int op, int64_t *out_k, void *out_v,
                          uint64_t *h_nout, void *d_ws, size_t ws_bytes) {
     Ws ws(d_ws, ws_bytes);
-    const uint64_t *sk, *sv;
-    hipError_t e = group_sort_u64((hipStream_t)stream, (const uint64_t *)in_k,
-                                  (const uint64_t *)in_v, n, 0, nullptr, ws, &sk, &sv);
-    if (e != hipSuccess) return e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP;
-    e = seg_reduce((hipStream_t)stream, sk, sv, n, op, (uint64_t *)out_k, out_v, h_nout, ws);
+    hipError_t e = group_sort_reduce((hipStream_t)stream, (const uint64_t *)in_k,
+                                     (const uint64_t *)in_v, n, op,
+                                     (uint64_t *)out_k, out_v, h_nout, ws);
+    if (e == hipErrorNotSupported) return VEGA_ERR_UNSUPPORTED;
     return e == hipSuccess ? VEGA_OK : (e == hipErrorOutOfMemory ? VEGA_ERR_NOMEM : VEGA_ERR_HIP);
 }
 

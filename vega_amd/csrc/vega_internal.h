@@ -66,7 +66,14 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 /* segmented aggregate over key-sorted rows: one output row per equal-key
  * run. op: VEGA_OP_*. Returns #segments in *h_nout (after stream sync). */
 hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t n,
-                      int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws);
+                      int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws,
+                      bool v_prezeroed = false);
+
+/* grouping sort + segmented aggregate, accumulator init overlapped on a
+ * side stream (the fast path reduce_by_key / group_count use) */
+hipError_t group_sort_reduce(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                             uint64_t n, int op, uint64_t *out_k, void *out_v,
+                             uint64_t *h_nout, Ws &ws);
 
 /* hash-mod partition scatter; h_counts on host after sync */
 hipError_t hash_partition(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,

@@ -1549,7 +1549,8 @@ __global__ void k_f64_seg_combine(const uint32_t *lead_seg, const double *lead_p
 }
 
 hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t n,
-                      int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws) {
+                      int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws,
+                      bool v_prezeroed) {
     if (n == 0) { *h_nout = 0; return hipSuccess; }
     if (n >= (1ULL << 32)) return hipErrorNotSupported; /* u32 head scan limit */
     uint32_t nb = nblocks_for(n);
@@ -1577,13 +1578,17 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     HIP_TRY(hipMemcpyAsync(&total, hc + nb, 4, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
 
-    /* init output accumulators */
-    if (op == 3 || op == 4) {
-        hipLaunchKernelGGL(k_fill_i64, dim3(2048), dim3(BLOCK), 0, s,
-                           (int64_t *)out_v, (uint64_t)total, op == 3 ? INT64_MAX : INT64_MIN);
-        HIP_TRY(hipGetLastError());
-    } else {
-        HIP_TRY(hipMemsetAsync(out_v, 0, (size_t)total * 8, s));
+    /* init output accumulators (skipped when the caller pre-initialized
+     * them overlapped with the sort, or for f64 — every f64 slot gets a
+     * plain store from its run's starting chunk) */
+    if (!v_prezeroed && op != 2) {
+        if (op == 3 || op == 4) {
+            hipLaunchKernelGGL(k_fill_i64, dim3(2048), dim3(BLOCK), 0, s,
+                               (int64_t *)out_v, (uint64_t)total, op == 3 ? INT64_MAX : INT64_MIN);
+            HIP_TRY(hipGetLastError());
+        } else {
+            HIP_TRY(hipMemsetAsync(out_v, 0, (size_t)total * 8, s));
+        }
     }
     {
         ProfScope ps("seg_emit", s);
@@ -1607,6 +1612,42 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     }
     *h_nout = total;
     return hipSuccess;
+}
+
+/* grouping sort + segmented aggregate with the accumulator init OVERLAPPED
+ * with the sort on a transient side stream (the n-slot init is ~1.2 ms of
+ * HBM writes at C1 that otherwise sits on the critical path). */
+hipError_t group_sort_reduce(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
+                             uint64_t n, int op, uint64_t *out_k, void *out_v,
+                             uint64_t *h_nout, Ws &ws) {
+    if (n == 0) { *h_nout = 0; return hipSuccess; }
+    hipStream_t s2 = nullptr;
+    hipEvent_t ev = nullptr;
+    bool prez = false;
+    if (op != 2 && n > (1u << 22)) { /* small n: stream setup not worth it */
+        if (hipStreamCreateWithFlags(&s2, hipStreamNonBlocking) == hipSuccess) {
+            if (op == 3 || op == 4) {
+                hipLaunchKernelGGL(k_fill_i64, dim3(2048), dim3(BLOCK), 0, s2,
+                                   (int64_t *)out_v, n, op == 3 ? INT64_MAX : INT64_MIN);
+            } else {
+                (void)hipMemsetAsync(out_v, 0, n * 8, s2);
+            }
+            if (hipEventCreateWithFlags(&ev, hipEventDisableTiming) == hipSuccess &&
+                hipEventRecord(ev, s2) == hipSuccess)
+                prez = true;
+        }
+    }
+    if (op == 2) prez = true;
+    const uint64_t *sk, *sv;
+    hipError_t e = group_sort_u64(s, in_k, in_v, n, 0, nullptr, ws, &sk, &sv);
+    if (e == hipSuccess) {
+        if (prez && ev) e = hipStreamWaitEvent(s, ev, 0);
+        if (e == hipSuccess)
+            e = seg_reduce(s, sk, sv, n, op, out_k, out_v, h_nout, ws, prez);
+    }
+    if (ev) (void)hipEventDestroy(ev);
+    if (s2) (void)hipStreamDestroy(s2);
+    return e;
 }
 
 /* ------------------------------------------------------------------ */
