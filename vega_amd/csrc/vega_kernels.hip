@@ -1614,40 +1614,18 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     return hipSuccess;
 }
 
-/* grouping sort + segmented aggregate with the accumulator init OVERLAPPED
- * with the sort on a transient side stream (the n-slot init is ~1.2 ms of
- * HBM writes at C1 that otherwise sits on the critical path). */
+/* grouping sort + segmented aggregate. (Overlapping the accumulator init
+ * with the sort on a side stream was measured a wash — the sort is itself
+ * HBM-bound, so the init just steals its bandwidth; f64 still skips the
+ * init entirely since its deterministic path plain-stores every slot.) */
 hipError_t group_sort_reduce(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
                              uint64_t n, int op, uint64_t *out_k, void *out_v,
                              uint64_t *h_nout, Ws &ws) {
     if (n == 0) { *h_nout = 0; return hipSuccess; }
-    hipStream_t s2 = nullptr;
-    hipEvent_t ev = nullptr;
-    bool prez = false;
-    if (op != 2 && n > (1u << 22)) { /* small n: stream setup not worth it */
-        if (hipStreamCreateWithFlags(&s2, hipStreamNonBlocking) == hipSuccess) {
-            if (op == 3 || op == 4) {
-                hipLaunchKernelGGL(k_fill_i64, dim3(2048), dim3(BLOCK), 0, s2,
-                                   (int64_t *)out_v, n, op == 3 ? INT64_MAX : INT64_MIN);
-            } else {
-                (void)hipMemsetAsync(out_v, 0, n * 8, s2);
-            }
-            if (hipEventCreateWithFlags(&ev, hipEventDisableTiming) == hipSuccess &&
-                hipEventRecord(ev, s2) == hipSuccess)
-                prez = true;
-        }
-    }
-    if (op == 2) prez = true;
     const uint64_t *sk, *sv;
     hipError_t e = group_sort_u64(s, in_k, in_v, n, 0, nullptr, ws, &sk, &sv);
-    if (e == hipSuccess) {
-        if (prez && ev) e = hipStreamWaitEvent(s, ev, 0);
-        if (e == hipSuccess)
-            e = seg_reduce(s, sk, sv, n, op, out_k, out_v, h_nout, ws, prez);
-    }
-    if (ev) (void)hipEventDestroy(ev);
-    if (s2) (void)hipStreamDestroy(s2);
-    return e;
+    if (e != hipSuccess) return e;
+    return seg_reduce(s, sk, sv, n, op, out_k, out_v, h_nout, ws, false);
 }
 
 /* ------------------------------------------------------------------ */
