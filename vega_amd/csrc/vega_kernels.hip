@@ -2082,27 +2082,56 @@ hipError_t cogroup_index(hipStream_t s, const int64_t *ka_u, uint64_t nka,
     return hipSuccess;
 }
 
+/* Both sides are sorted, so a block's contiguous 4096-row span of A maps to
+ * a narrow window of B: two block-level searches bound it, then every
+ * per-row search runs inside the (L2-hot) window — ~12 probes instead of
+ * ~29 cold ones per row at the C4 shape. */
 __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
                              uint64_t nb, int mode, uint32_t *counts,
                              uint32_t *b_lo) {
-    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < na; i += stride) {
-        int64_t k = ak[i];
-        /* lower bound */
-        uint64_t lo = 0, hi = nb;
-        while (lo < hi) {
-            uint64_t m = (lo + hi) >> 1;
-            if (join_less(mode, bk[m], k)) lo = m + 1; else hi = m;
+    __shared__ uint64_t s_lo, s_hi;
+    uint64_t nspans = (na + TILE - 1) / TILE;
+    for (uint64_t sp = blockIdx.x; sp < nspans; sp += gridDim.x) {
+        uint64_t a0 = sp * TILE;
+        uint64_t a1 = (na - a0 < TILE) ? na : a0 + TILE;
+        if (threadIdx.x == 0) {
+            int64_t kf = ak[a0];
+            uint64_t lo = 0, hi = nb;
+            while (lo < hi) {
+                uint64_t m = (lo + hi) >> 1;
+                if (join_less(mode, bk[m], kf)) lo = m + 1; else hi = m;
+            }
+            s_lo = lo;
+        } else if (threadIdx.x == 64) {
+            int64_t kl = ak[a1 - 1];
+            uint64_t lo = 0, hi = nb;
+            while (lo < hi) {
+                uint64_t m = (lo + hi) >> 1;
+                if (!join_less(mode, kl, bk[m])) lo = m + 1; else hi = m;
+            }
+            s_hi = lo;
         }
-        uint64_t lb = lo;
-        /* upper bound */
-        hi = nb;
-        while (lo < hi) {
-            uint64_t m = (lo + hi) >> 1;
-            if (!join_less(mode, k, bk[m])) lo = m + 1; else hi = m;
+        __syncthreads();
+        const uint64_t wlo = s_lo, whi = s_hi;
+        __syncthreads(); /* s_lo/s_hi free for the next span */
+        for (uint64_t i = a0 + threadIdx.x; i < a1; i += blockDim.x) {
+            int64_t k = ak[i];
+            /* lower bound within the span window */
+            uint64_t lo = wlo, hi = whi;
+            while (lo < hi) {
+                uint64_t m = (lo + hi) >> 1;
+                if (join_less(mode, bk[m], k)) lo = m + 1; else hi = m;
+            }
+            uint64_t lb = lo;
+            /* upper bound */
+            hi = whi;
+            while (lo < hi) {
+                uint64_t m = (lo + hi) >> 1;
+                if (!join_less(mode, k, bk[m])) lo = m + 1; else hi = m;
+            }
+            counts[i] = (uint32_t)(lo - lb);
+            b_lo[i] = (uint32_t)lb;
         }
-        counts[i] = (uint32_t)(lo - lb);
-        b_lo[i] = (uint32_t)lb;
     }
 }
 

@@ -98,10 +98,17 @@ def partition_range_cpu(k, v, splitters):
 
 
 def all_to_all_kv(send_k, send_v, send_counts, group=None):
-    """Exchange bucket-contiguous (k, v) rows: counts all-to-all, then
-    payload all-to-all-v. Tensors stay on their device (CUDA -> RCCL/xGMI,
-    CPU -> gloo)."""
+    """Exchange bucket-contiguous (k, v) rows: counts all-to-all (replacing
+    the MapOutputTracker), then the payload exchange as grouped P2P
+    send/recv (RCCL over xGMI on CUDA, gloo on CPU — same code path).
+
+    The rank's OWN bucket never touches the collective: it moves with a
+    device copy (all_to_all_single was measured moving the self-shard at
+    ~8 GB/s — on the 8-GPU node 1/8 of every exchange would have crawled
+    while the 7 xGMI links idled)."""
     dev = send_k.device
+    world = dist.get_world_size(group=group)
+    rank = dist.get_rank(group=group)
     sc = torch.as_tensor(send_counts, dtype=torch.int64, device=dev)
     rc = torch.empty_like(sc)
     dist.all_to_all_single(rc, sc, group=group)
@@ -110,10 +117,31 @@ def all_to_all_kv(send_k, send_v, send_counts, group=None):
     nrecv = sum(out_splits)
     recv_k = torch.empty(nrecv, dtype=send_k.dtype, device=dev)
     recv_v = torch.empty(nrecv, dtype=send_v.dtype, device=dev)
-    dist.all_to_all_single(recv_k, send_k[:sum(in_splits)],
-                           output_split_sizes=out_splits,
-                           input_split_sizes=in_splits, group=group)
-    dist.all_to_all_single(recv_v, send_v[:sum(in_splits)],
-                           output_split_sizes=out_splits,
-                           input_split_sizes=in_splits, group=group)
+    soff = [0] * world
+    roff = [0] * world
+    for p in range(1, world):
+        soff[p] = soff[p - 1] + in_splits[p - 1]
+        roff[p] = roff[p - 1] + out_splits[p - 1]
+    ops = []
+    for p in range(world):
+        if p == rank:
+            continue
+        if in_splits[p]:
+            ops.append(dist.P2POp(dist.isend, send_k[soff[p]:soff[p] + in_splits[p]],
+                                  p, group=group))
+            ops.append(dist.P2POp(dist.isend, send_v[soff[p]:soff[p] + in_splits[p]],
+                                  p, group=group))
+        if out_splits[p]:
+            ops.append(dist.P2POp(dist.irecv, recv_k[roff[p]:roff[p] + out_splits[p]],
+                                  p, group=group))
+            ops.append(dist.P2POp(dist.irecv, recv_v[roff[p]:roff[p] + out_splits[p]],
+                                  p, group=group))
+    if in_splits[rank]:  # self bucket: plain device copy, off the wire
+        recv_k[roff[rank]:roff[rank] + in_splits[rank]].copy_(
+            send_k[soff[rank]:soff[rank] + in_splits[rank]])
+        recv_v[roff[rank]:roff[rank] + in_splits[rank]].copy_(
+            send_v[soff[rank]:soff[rank] + in_splits[rank]])
+    if ops:
+        for req in dist.batch_isend_irecv(ops):
+            req.wait()
     return recv_k, recv_v
