@@ -8,9 +8,14 @@ python -c "import __graft_entry__ as g; g.smoke()"           # tiny reduce vs or
 ./vega_amd/host/vega_cli selftest                            # C++ host mirror goldens
 python bench.py                                              # the contract line (C1, 1e9)
 python bench.py --op group_count --dist zipf --steps 4 --warmup 1 --no-cpu-baseline  # C2
-python bench.py --op sort  --rows 500000000 --steps 3 --warmup 1 --no-cpu-baseline   # C3 shape
+python bench.py --op sort  --rows 2000000000 --steps 3 --warmup 1 --no-cpu-baseline  # C3 official total size, 1 GPU
 python bench.py --op join  --rows 500000000 --steps 3 --warmup 1 --no-cpu-baseline   # C4 shape
-python bench.py --dtype f64 --steps 5 --warmup 2 --no-cpu-baseline                   # C1 f64 variant
+python bench.py --dtype f64 --steps 5 --warmup 2 --no-cpu-baseline                   # C1 f64 variant (deterministic)
+# RCCL exchange path on hardware (1 GPU, world 1):
+python -m torch.distributed.run --nnodes=1 --nproc-per-node 1 --master-addr 127.0.0.1 \
+  --master-port 29513 bench.py --force-dist --steps 5 --warmup 2 --no-cpu-baseline
+# scatter phase breakdown (s_memtime instrumentation):
+VEGA_PHASE_PROF=1 python tools/phase_prof.py --rows 200000000
 # profiling (PMC in separate runs; never combined with trace domains):
 export TMPDIR=/tmp
 (cd /tmp && rocprofv3 --kernel-trace --stats -d $OLDPWD/gpurun_out/kt -o kt -- python $OLDPWD/bench.py --rows 200000000 --steps 2 --warmup 1 --no-cpu-baseline)
