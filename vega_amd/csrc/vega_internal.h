@@ -131,8 +131,9 @@ hipError_t cogroup_index(hipStream_t s, const int64_t *ka_u, uint64_t nka,
 
 size_t ws_bytes_for(uint64_t n);
 
-/* diagnostic phase-cycle buffer (VEGA_PHASE_PROF=1), else nullptr */
+/* diagnostic phase-cycle buffer (VEGA_PHASE_PROF=1|2), else nullptr */
 unsigned long long *phase_prof_buf();
+int phase_prof_mode(); /* 0 off, 1 phases, 2 phases + walk counters */
 
 } // namespace vega
 

@@ -613,7 +613,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
          * lines per publish) */
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            (unsigned long long)cnt | OSW_ST_AGG |
-                               ((unsigned long long)ptag << 56),
+                               ((unsigned long long)(ptag & 0x3F) << 56),
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         /* second level: add this tile's count into its 32-tile GROUP slot
          * (arrivals in the high bits, sum in the low) — one relaxed 8-byte
@@ -666,7 +666,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 unsigned long long dd4[4] = {d0, d1, d2, d3};
                 for (int q = 0; q < navail; ++q) {
                     unsigned long long st =
-                        (OSW_TAG(dd4[q]) == (uint32_t)ptag) ? (dd4[q] >> 62) : 0;
+                        (OSW_TAG(dd4[q]) == (uint32_t)(ptag & 0x3F)) ? (dd4[q] >> 62) : 0;
                     if (st == 2) { excl_tiles += dd4[q] & OSW_CNT_MASK; done = true; break; }
                     if (st == 1) { excl_tiles += dd4[q] & OSW_CNT_MASK; j--; continue; }
                     stall = true;
@@ -695,7 +695,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                     unsigned long long ge = __hip_atomic_load(
                         col + (((uint64_t)g + 1) * OSW_GRP - 1) * 256, __ATOMIC_RELAXED,
                         __HIP_MEMORY_SCOPE_AGENT);
-                    if ((ge >> 62) == 2 && OSW_TAG(ge) == (uint32_t)ptag) {
+                    if ((ge >> 62) == 2 && OSW_TAG(ge) == (uint32_t)(ptag & 0x3F)) {
                         /* INC: covers [0, (g+1)*GRP-1] */
                         excl_tiles += ge & OSW_CNT_MASK;
                         break;
@@ -715,14 +715,15 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                     niter++;
                 }
             }
-            if (phc) { /* diagnostic: walk iterations vs publish stalls */
+            if (phc && (ptag & 0x100)) { /* walk counters: mode 2 only (they
+                 * serialize on two hot addresses and distort the phases) */
                 atomicAdd(&phc[6], (unsigned long long)niter);
                 atomicAdd(&phc[7], (unsigned long long)nstall);
             }
         }
         __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
                            ((excl_tiles + cnt) | OSW_ST_INC) |
-                               ((unsigned long long)ptag << 56),
+                               ((unsigned long long)(ptag & 0x3F) << 56),
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
     }
@@ -779,18 +780,21 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
 /* diagnostic per-phase cycle sums (VEGA_PHASE_PROF=1): lazily allocated,
  * read/reset via vega_phase_prof_read() */
 static unsigned long long *g_phase_buf = nullptr;
+static int g_phase_mode = 0; /* 1 = phases only, 2 = + walk counters */
 unsigned long long *phase_prof_buf() {
     static bool checked = false;
     if (!checked) {
         checked = true;
         const char *e = getenv("VEGA_PHASE_PROF");
-        if (e && e[0] == '1') {
+        if (e && (e[0] == '1' || e[0] == '2')) {
+            g_phase_mode = e[0] - '0';
             (void)hipMalloc(&g_phase_buf, 8 * 8);
             (void)hipMemset(g_phase_buf, 0, 8 * 8);
         }
     }
     return g_phase_buf;
 }
+int phase_prof_mode() { return g_phase_mode; }
 
 template <class DF>
 static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
@@ -801,6 +805,7 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
                                    bool has_vals, bool in_pk, bool out_pk,
                                    DF df, const char *prof_name) {
     unsigned long long *phc = phase_prof_buf();
+    if (phase_prof_mode() == 2) ptag |= 0x100; /* enable walk counters */
     uint32_t nb = nblocks_for(n);
     /* desc is zeroed once per SORT call (pass tags invalidate stale words) */
     HIP_TRY(hipMemsetAsync(ff, 0, (size_t)((nb + OSW_GRP - 1) / OSW_GRP) * 256 * 8, s));
