@@ -25,8 +25,8 @@ def main():
     ap.add_argument("--reps", type=int, default=3)
     ap.add_argument("--op", default="reduce", choices=["reduce", "sort"])
     args = ap.parse_args()
-    if os.environ.get("VEGA_PHASE_PROF") != "1":
-        print("set VEGA_PHASE_PROF=1", file=sys.stderr)
+    if os.environ.get("VEGA_PHASE_PROF") not in ("1", "2"):
+        print("set VEGA_PHASE_PROF=1 (phases) or 2 (+walk counters)", file=sys.stderr)
         sys.exit(2)
     import torch
     from vega_amd import gpu

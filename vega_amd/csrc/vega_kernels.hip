@@ -673,7 +673,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                     break;
                 }
                 if (done) break;
-                if (stall) { nstall++; __builtin_amdgcn_s_sleep(1); }
+                if (stall) nstall++; /* busy retry: the reload IS the backoff */
                 niter++;
             }
             /* 2) whole groups below: one load consumes 32 ranked tiles;
@@ -711,7 +711,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                         stall = true;
                         break;
                     }
-                    if (stall) { nstall++; __builtin_amdgcn_s_sleep(1); }
+                    if (stall) nstall++; /* busy retry */
                     niter++;
                 }
             }
