@@ -517,11 +517,11 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     ulonglong2 *spk = (ulonglong2 *)smem;                       /* TILE 16B */
     uint64_t *sk = (uint64_t *)smem;                            /* no-vals */
-    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 16 : 8) * (size_t)TILE);
-    uint32_t *whist = hist + 256;                               /* SW*256 */
-    uint32_t *wsc = whist + SW * 256;                           /* 8 */
-    uint32_t *vbp = wsc + 8;                                    /* 8 */
-    uint32_t *tilebase = vbp + 8;                               /* 256 */
+    uint32_t *hist = (uint32_t *)(smem + (HAS_VALS ? 16 : 8) * (size_t)TILE); /* 2x256 */
+    uint32_t *whist = hist + 2 * 256;                           /* SW*256 */
+    uint32_t *wsc = whist + SW * 256;                           /* SW */
+    uint32_t *vbp = wsc + SW;                                   /* 8 */
+    uint32_t *tilebase = vbp + 8;                               /* 2x256 */
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t lower = ((uint64_t)1 << lane) - 1;
@@ -533,41 +533,31 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
         tprev = tnow;                                                          \
     }
 
-    if (t == 0) vbp[0] = atomicAdd(ticket, 1u); /* scheduling-ordered tile id */
-    for (int i = t; i < SW * 256; i += SB) whist[i] = 0;
-    __syncthreads();
-    const uint32_t vb = vbp[0];
-    const uint64_t tbase = (uint64_t)vb * TILE;
-    const uint32_t tile_n = (uint32_t)((n - tbase < TILE) ? (n - tbase) : TILE);
-
-    /* prefetch the wave's whole 512-row chunk (independent loads in flight
-     * together: ONE memory latency per chunk, not one per round) */
+    /* ---- SOFTWARE PIPELINE ACROSS TILES ----
+     * A tile's lookback is DEFERRED until after the NEXT tile's rank and
+     * publish: by then the current generation's rank stragglers have
+     * published their AGG/group counts, so the walk resolves without the
+     * convoy stalls the flat schedule paid (~60% of all wave cycles,
+     * profiles/r02_phase_prof_ladder.txt). Blocks are quasi-persistent
+     * (grid = min(nb, 512)) and pull ticket-ordered tiles until exhausted;
+     * hist/tilebase are double-buffered per slot while the 64 KB row
+     * staging stays single (writeout N frees it before reorder N+1). */
     uint64_t kk[SIPT], vv[SIPT];
     uint32_t rank[SIPT];
     uint16_t dd[SIPT];
-    const uint64_t chunk_g = tbase + (uint64_t)w * (64 * SIPT) + lane;
-    const bool chunk_full = tbase + ((uint64_t)w + 1) * (64 * SIPT) <= n;
-    if (chunk_full) {
+    uint32_t tn_cur = 0;
+
+    auto stage_rank = [&](uint32_t vb, int slot) {
+        const uint64_t tbase = (uint64_t)vb * TILE;
+        tn_cur = (uint32_t)((n - tbase < TILE) ? (n - tbase) : TILE);
+        const uint64_t chunk_g = tbase + (uint64_t)w * (64 * SIPT) + lane;
+        const bool chunk_full = tbase + ((uint64_t)w + 1) * (64 * SIPT) <= n;
+        /* prefetch the wave's whole 512-row chunk (independent loads in
+         * flight together: ONE memory latency per chunk) */
+        if (chunk_full) {
 #pragma unroll
-        for (int r = 0; r < SIPT; ++r) {
-            uint64_t idx = chunk_g + (uint64_t)r * 64;
-            if (IN_PK) {
-                ulonglong2 kv = ((const ulonglong2 *)in_k)[idx];
-                kk[r] = kv.x;
-                vv[r] = kv.y;
-            } else {
-                kk[r] = in_k[idx];
-                if (HAS_VALS) vv[r] = in_v[idx];
-            }
-        }
-    } else {
-#pragma unroll
-        for (int r = 0; r < SIPT; ++r) {
-            uint64_t idx = chunk_g + (uint64_t)r * 64;
-            bool valid = idx < n;
-            kk[r] = 0;
-            vv[r] = 0;
-            if (valid) {
+            for (int r = 0; r < SIPT; ++r) {
+                uint64_t idx = chunk_g + (uint64_t)r * 64;
                 if (IN_PK) {
                     ulonglong2 kv = ((const ulonglong2 *)in_k)[idx];
                     kk[r] = kv.x;
@@ -577,72 +567,93 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                     if (HAS_VALS) vv[r] = in_v[idx];
                 }
             }
-        }
-    }
-    VEGA_PHASE_MARK(0)
-    /* ranking: register-only rounds over the prefetched chunk */
+        } else {
 #pragma unroll
-    for (int r = 0; r < SIPT; ++r) {
-        bool valid = chunk_g + (uint64_t)r * 64 < n;
-        uint32_t d = valid ? df(kk[r]) : 0;
-        uint64_t m = wave_match8(d, valid);
-        int leader_lane = (int)__ffsll((unsigned long long)m) - 1;
-        if (leader_lane < 0) leader_lane = 0;
-        uint32_t base = 0;
-        if (valid && lane == leader_lane)
-            base = atomicAdd(&whist[w * 256 + d], (uint32_t)__popcll(m));
-        base = __shfl(base, leader_lane);
-        dd[r] = (uint16_t)d;
-        rank[r] = base + (uint32_t)__popcll(m & lower);
-    }
-    __syncthreads();
-    VEGA_PHASE_MARK(1)
+            for (int r = 0; r < SIPT; ++r) {
+                uint64_t idx = chunk_g + (uint64_t)r * 64;
+                bool valid = idx < n;
+                kk[r] = 0;
+                vv[r] = 0;
+                if (valid) {
+                    if (IN_PK) {
+                        ulonglong2 kv = ((const ulonglong2 *)in_k)[idx];
+                        kk[r] = kv.x;
+                        vv[r] = kv.y;
+                    } else {
+                        kk[r] = in_k[idx];
+                        if (HAS_VALS) vv[r] = in_v[idx];
+                    }
+                }
+            }
+        }
+        VEGA_PHASE_MARK(0)
+        /* ranking: register-only rounds over the prefetched chunk */
+#pragma unroll
+        for (int r = 0; r < SIPT; ++r) {
+            bool valid = chunk_g + (uint64_t)r * 64 < n;
+            uint32_t d = valid ? df(kk[r]) : 0;
+            uint64_t m = wave_match8(d, valid);
+            int leader_lane = (int)__ffsll((unsigned long long)m) - 1;
+            if (leader_lane < 0) leader_lane = 0;
+            uint32_t base = 0;
+            if (valid && lane == leader_lane)
+                base = atomicAdd(&whist[w * 256 + d], (uint32_t)__popcll(m));
+            base = __shfl(base, leader_lane);
+            dd[r] = (uint16_t)d;
+            rank[r] = base + (uint32_t)__popcll(m & lower);
+        }
+        __syncthreads();
+        VEGA_PHASE_MARK(1)
 
-    /* digit threads (t<256): counts -> publish AGG, local starts, per-wave
-     * offsets, lookback */
-    uint32_t cw[SW];
-    uint32_t cnt = 0;
-    if (t < 256) {
+        /* digit threads: counts -> publish AGG + group add, local starts,
+         * per-wave offsets. NO lookback here — it runs one tile later. */
+        uint32_t cw[SW];
+        uint32_t cnt = 0;
+        if (t < 256) {
 #pragma unroll
-        for (int wv = 0; wv < SW; ++wv) {
-            cw[wv] = whist[wv * 256 + t];
-            cnt += cw[wv];
-        }
-        /* tile-major descriptors: a tile's 256 publishes are one contiguous
-         * 2 KB burst (digit-major measured 3.6x slower — 256 scattered
-         * lines per publish) */
-        __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
-                           (unsigned long long)cnt | OSW_ST_AGG |
-                               ((unsigned long long)(ptag & 0x3F) << 56),
-                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        /* second level: add this tile's count into its 32-tile GROUP slot
-         * (arrivals in the high bits, sum in the low) — one relaxed 8-byte
-         * atomicAdd granule. A full group (arrivals == 32) is consumable
-         * with ONE load, and group sums need only the RANK front — they
-         * never wait on predecessors' lookbacks, which is what breaks the
-         * walk-length/retire-front feedback the flat walk suffered from. */
-        __hip_atomic_fetch_add((gdesc_t *)&ff[((uint64_t)vb >> OSW_GRP_LG) * 256 + t],
-                               (1ULL << 42) | (unsigned long long)cnt,
+            for (int wv = 0; wv < SW; ++wv) {
+                cw[wv] = whist[wv * 256 + t];
+                cnt += cw[wv];
+            }
+            /* tile-major descriptors: a tile's 256 publishes are one
+             * contiguous 2 KB burst */
+            __hip_atomic_store((gdesc_t *)&desc[(uint64_t)vb * 256 + t],
+                               (unsigned long long)cnt | OSW_ST_AGG |
+                                   ((unsigned long long)(ptag & 0x3F) << 56),
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    }
-    uint32_t inc = cnt;
-    for (int off = 1; off < 64; off <<= 1) {
-        uint32_t u = __shfl_up(inc, off);
-        if (lane >= off) inc += u;
-    }
-    if (lane == 63) wsc[w] = inc;
-    __syncthreads();
-    if (t < 256) {
-        uint32_t excl = inc - cnt;
-        for (int i = 0; i < w; ++i) excl += wsc[i];
-        hist[t] = excl;
-        uint32_t run = 0;
+            /* second level: per-group (arrivals | sum) — one relaxed 8-byte
+             * atomicAdd granule; group sums follow the RANK front, so walks
+             * never wait on predecessors' (deferred) lookbacks */
+            __hip_atomic_fetch_add(
+                (gdesc_t *)&ff[((uint64_t)vb >> OSW_GRP_LG) * 256 + t],
+                (1ULL << 42) | (unsigned long long)cnt,
+                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+        uint32_t inc = cnt;
+        for (int off = 1; off < 64; off <<= 1) {
+            uint32_t u = __shfl_up(inc, off);
+            if (lane >= off) inc += u;
+        }
+        if (lane == 63) wsc[w] = inc;
+        __syncthreads();
+        if (t < 256) {
+            uint32_t excl = inc - cnt;
+            for (int i = 0; i < w; ++i) excl += wsc[i];
+            hist[slot * 256 + t] = excl;
+            uint32_t run = 0;
 #pragma unroll
-        for (int wv = 0; wv < SW; ++wv) {
-            whist[wv * 256 + t] = run;
-            run += cw[wv];
+            for (int wv = 0; wv < SW; ++wv) {
+                whist[wv * 256 + t] = run;
+                run += cw[wv];
+            }
         }
         VEGA_PHASE_MARK(2)
+    };
+
+    auto stage_lookback = [&](uint32_t vb, int slot, uint32_t tn) {
+        if (t >= 256) return;
+        uint32_t excl_local = hist[slot * 256 + t];
+        uint32_t cnt = ((t < 255) ? hist[slot * 256 + t + 1] : tn) - excl_local;
         unsigned long long excl_tiles = 0;
         if (vb > 0) {
             gdesc_t *col = (gdesc_t *)(desc + t);
@@ -653,8 +664,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
             int64_t gb_lo = (int64_t)(vb & ~(uint32_t)(OSW_GRP - 1));
             int64_t j = (int64_t)vb - 1;
             while (j >= gb_lo) {
-                /* bounded spin: a lost predecessor can never wedge the GPU —
-                 * set the abort flag, bail, and let the host fail loudly */
+                /* bounded spin: a lost predecessor can never wedge the GPU */
                 if (++spins > (1u << 26)) { *d_abort = 1; done = true; break; }
                 unsigned long long d0, d1 = 0, d2 = 0, d3 = 0;
                 int navail = (j - gb_lo >= 3) ? 4 : (int)(j - gb_lo + 1);
@@ -676,16 +686,12 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 if (stall) nstall++; /* busy retry: the reload IS the backoff */
                 niter++;
             }
-            /* 2) whole groups below: one load consumes 32 ranked tiles;
-             * the group's last-tile descriptor doubles as the deep-history
-             * INC shortcut */
+            /* 2) whole groups below: one load consumes OSW_GRP ranked tiles;
+             * the group's last-tile descriptor doubles as the INC shortcut */
             if (!done && vb >= OSW_GRP) {
                 int64_t g = (int64_t)(vb >> OSW_GRP_LG) - 1;
                 while (g >= 0) {
                     if (++spins > (1u << 26)) { *d_abort = 1; break; }
-                    /* probe 4 groups per round trip: each iteration issues
-                     * the 4 group sums + the eldest's INC-shortcut
-                     * descriptor as independent loads */
                     int ga = (g >= 3) ? 4 : (int)(g + 1);
                     unsigned long long g0, g1 = 0, g2 = 0, g3 = 0;
                     g0 = __hip_atomic_load(gcol + (uint64_t)g * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -696,14 +702,13 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                         col + (((uint64_t)g + 1) * OSW_GRP - 1) * 256, __ATOMIC_RELAXED,
                         __HIP_MEMORY_SCOPE_AGENT);
                     if ((ge >> 62) == 2 && OSW_TAG(ge) == (uint32_t)(ptag & 0x3F)) {
-                        /* INC: covers [0, (g+1)*GRP-1] */
-                        excl_tiles += ge & OSW_CNT_MASK;
+                        excl_tiles += ge & OSW_CNT_MASK; /* covers [0, (g+1)*GRP-1] */
                         break;
                     }
                     unsigned long long gg4[4] = {g0, g1, g2, g3};
                     bool stall = false;
                     for (int q = 0; q < ga; ++q) {
-                        if ((gg4[q] >> 42) == OSW_GRP) { /* group fully ranked */
+                        if ((gg4[q] >> 42) == OSW_GRP) {
                             excl_tiles += gg4[q] & ((1ULL << 42) - 1);
                             g--;
                             continue;
@@ -711,12 +716,11 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                         stall = true;
                         break;
                     }
-                    if (stall) nstall++; /* busy retry */
+                    if (stall) nstall++;
                     niter++;
                 }
             }
-            if (phc && (ptag & 0x100)) { /* walk counters: mode 2 only (they
-                 * serialize on two hot addresses and distort the phases) */
+            if (phc && (ptag & 0x100)) { /* walk counters: mode 2 only */
                 atomicAdd(&phc[6], (unsigned long long)niter);
                 atomicAdd(&phc[7], (unsigned long long)nstall);
             }
@@ -725,55 +729,91 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                            ((excl_tiles + cnt) | OSW_ST_INC) |
                                ((unsigned long long)(ptag & 0x3F) << 56),
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        tilebase[t] = gbase[t] + (uint32_t)excl_tiles;
-    }
-    __syncthreads();
-    VEGA_PHASE_MARK(3)
+        tilebase[slot * 256 + t] = gbase[t] + (uint32_t)excl_tiles;
+    };
 
-    /* reorder into LDS at the stable tile-local position */
-    {
+    auto stage_reorder = [&](int slot, uint32_t tn) {
         uint32_t chunk0 = (uint32_t)w * (64 * SIPT);
-        uint32_t chunk_n = tile_n > chunk0 ? tile_n - chunk0 : 0;
+        uint32_t chunk_n = tn > chunk0 ? tn - chunk0 : 0;
 #pragma unroll
         for (int r = 0; r < SIPT; ++r) {
             uint32_t local = (uint32_t)r * 64 + lane;
             if (local < chunk_n) {
                 uint32_t d = dd[r];
-                uint32_t pos = hist[d] + whist[w * 256 + d] + rank[r];
+                uint32_t pos = hist[slot * 256 + d] + whist[w * 256 + d] + rank[r];
                 if (HAS_VALS) spk[pos] = make_ulonglong2(kk[r], vv[r]);
                 else sk[pos] = kk[r];
             }
         }
-    }
-    __syncthreads();
-    VEGA_PHASE_MARK(4)
+        VEGA_PHASE_MARK(4)
+    };
 
-    /* write out: digit-contiguous global writes */
-    for (uint32_t p = t; p < tile_n; p += SB) {
-        uint64_t k, v = 0;
-        if (HAS_VALS) {
-            ulonglong2 kv = spk[p];
-            k = kv.x;
-            v = kv.y;
-        } else {
-            k = sk[p];
+    auto stage_writeout = [&](int slot, uint32_t tn) {
+        for (uint32_t p = t; p < tn; p += SB) {
+            uint64_t k, v = 0;
+            if (HAS_VALS) {
+                ulonglong2 kv = spk[p];
+                k = kv.x;
+                v = kv.y;
+            } else {
+                k = sk[p];
+            }
+            uint32_t d = df(k);
+            uint64_t gpos = (uint64_t)tilebase[slot * 256 + d] + (p - hist[slot * 256 + d]);
+            /* plain stores: L2 write-combining of the 16-B digit-run stores
+             * is load-bearing (nontemporal regressed 33%) */
+            if (OUT_PK) {
+                ((ulonglong2 *)out_k)[gpos] = make_ulonglong2(k, v);
+            } else {
+                out_k[gpos] = k;
+                if (HAS_VALS) out_v[gpos] = v;
+                if (h32_out) /* low hash bits for the grouping cleanup */
+                    h32_out[gpos] = (uint32_t)vega_hash_u64(k);
+            }
         }
-        uint32_t d = df(k);
-        uint64_t gpos = (uint64_t)tilebase[d] + (p - hist[d]);
-        /* plain stores: L2 write-combining of the 16-B digit-run stores is
-         * load-bearing — nontemporal stores regressed the pass 33% (partial
-         * 64-B sectors straight to DRAM) */
-        if (OUT_PK) {
-            ((ulonglong2 *)out_k)[gpos] = make_ulonglong2(k, v);
-        } else {
-            out_k[gpos] = k;
-            if (HAS_VALS) out_v[gpos] = v;
-            if (h32_out) /* low hash bits for the grouping cleanup (4 B/row
-                            beats its re-hashing 16 B/row) */
-                h32_out[gpos] = (uint32_t)vega_hash_u64(k);
+        VEGA_PHASE_MARK(5)
+    };
+
+    /* prologue: tile A through rank+reorder */
+    for (int i = t; i < SW * 256; i += SB) whist[i] = 0;
+    if (t == 0) vbp[0] = atomicAdd(ticket, 1u);
+    __syncthreads();
+    uint32_t vbA = vbp[0];
+    if (vbA >= nblocks) return; /* block got no work (uniform) */
+    int slotA = 0;
+    stage_rank(vbA, 0);
+    uint32_t tnA = tn_cur;
+    __syncthreads();
+    stage_reorder(0, tnA);
+    __syncthreads();
+    for (int i = t; i < SW * 256; i += SB) whist[i] = 0;
+    if (t == 0) vbp[1] = atomicAdd(ticket, 1u);
+    __syncthreads();
+
+    while (true) {
+        int slotB = 1 - slotA;
+        uint32_t vbB = vbp[slotB];
+        bool haveB = vbB < nblocks; /* uniform */
+        uint32_t tnB = 0;
+        if (haveB) {
+            stage_rank(vbB, slotB);
+            tnB = tn_cur;
         }
+        stage_lookback(vbA, slotA, tnA); /* deferred past B's rank+publish */
+        __syncthreads();
+        VEGA_PHASE_MARK(3)
+        stage_writeout(slotA, tnA);
+        if (!haveB) return;
+        __syncthreads(); /* row staging free for B */
+        stage_reorder(slotB, tnB);
+        __syncthreads();
+        for (int i = t; i < SW * 256; i += SB) whist[i] = 0;
+        if (t == 0) vbp[slotA] = atomicAdd(ticket, 1u);
+        __syncthreads();
+        vbA = vbB;
+        tnA = tnB;
+        slotA = slotB;
     }
-    VEGA_PHASE_MARK(5)
 #undef VEGA_PHASE_MARK
 }
 
@@ -811,21 +851,23 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     HIP_TRY(hipMemsetAsync(ff, 0, (size_t)((nb + OSW_GRP - 1) / OSW_GRP) * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
-    size_t sh = (has_vals ? 16 : 8) * (size_t)TILE + (256 + 8 * 256 + 8 + 8 + 256) * 4;
+    size_t sh = (has_vals ? 16 : 8) * (size_t)TILE +
+                (2 * 256 + 8 * 256 + 8 + 8 + 2 * 256) * 4;
+    uint32_t grid = nb < 512 ? nb : 512; /* quasi-persistent: tiles pulled by ticket */
     if (!has_vals) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(nb), dim3(512), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, false, false, false>), dim3(grid), dim3(512), sh, s,
                            in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, ptag, df);
     } else if (!in_pk && !out_pk) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(nb), dim3(512), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, false>), dim3(grid), dim3(512), sh, s,
                            in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, ptag, df);
     } else if (!in_pk && out_pk) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(nb), dim3(512), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, false, true>), dim3(grid), dim3(512), sh, s,
                            in_k, in_v, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, ptag, df);
     } else if (in_pk && out_pk) {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(nb), dim3(512), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, true>), dim3(grid), dim3(512), sh, s,
                            in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, nullptr, h32_out, d_abort, ff, phc, ptag, df);
     } else {
-        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(nb), dim3(512), sh, s,
+        hipLaunchKernelGGL((k_scatter_osw<DF, true, true, false>), dim3(grid), dim3(512), sh, s,
                            in_k, nullptr, n, nb, gbase_d, desc, ticket, out_k, out_v, h32_out, d_abort, ff, phc, ptag, df);
     }
     return hipGetLastError();
