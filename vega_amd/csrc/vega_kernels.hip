@@ -490,6 +490,8 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(
 
 #define OSW_GRP_LG 4
 #define OSW_GRP (1 << OSW_GRP_LG)
+#define OSW_SUP_LG 8
+#define OSW_SUP (1 << OSW_SUP_LG)
 #define OSW_ST_AGG (1ULL << 62)
 #define OSW_ST_INC (2ULL << 62)
 #define OSW_CNT_MASK ((1ULL << 56) - 1)
@@ -628,6 +630,14 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 (gdesc_t *)&ff[((uint64_t)vb >> OSW_GRP_LG) * 256 + t],
                 (1ULL << 42) | (unsigned long long)cnt,
                 __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            /* third level: 256-tile SUPER slot — the deferred-INC front sits
+             * ~a pipeline stage (~512 tiles) back, so the deep walk crosses
+             * it in one or two super probes instead of ~32 group loads */
+            __hip_atomic_fetch_add(
+                (gdesc_t *)&ff[((uint64_t)((nblocks + OSW_GRP - 1) >> OSW_GRP_LG) +
+                               ((uint64_t)vb >> OSW_SUP_LG)) * 256 + t],
+                (1ULL << 42) | (unsigned long long)cnt,
+                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         }
         uint32_t inc = cnt;
         for (int off = 1; off < 64; off <<= 1) {
@@ -686,13 +696,15 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                 if (stall) nstall++; /* busy retry: the reload IS the backoff */
                 niter++;
             }
-            /* 2) whole groups below: one load consumes OSW_GRP ranked tiles;
-             * the group's last-tile descriptor doubles as the INC shortcut */
+            /* 2) whole groups below, down to the own 256-tile super
+             * boundary; the group's last-tile descriptor doubles as the
+             * INC shortcut */
+            int64_t sup_lo_g = (int64_t)((vb & ~(uint32_t)(OSW_SUP - 1)) >> OSW_GRP_LG);
             if (!done && vb >= OSW_GRP) {
                 int64_t g = (int64_t)(vb >> OSW_GRP_LG) - 1;
-                while (g >= 0) {
+                while (g >= sup_lo_g) {
                     if (++spins > (1u << 26)) { *d_abort = 1; break; }
-                    int ga = (g >= 3) ? 4 : (int)(g + 1);
+                    int ga = (g - sup_lo_g >= 3) ? 4 : (int)(g - sup_lo_g + 1);
                     unsigned long long g0, g1 = 0, g2 = 0, g3 = 0;
                     g0 = __hip_atomic_load(gcol + (uint64_t)g * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                     if (ga > 1) g1 = __hip_atomic_load(gcol + (uint64_t)(g - 1) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -711,6 +723,43 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                         if ((gg4[q] >> 42) == OSW_GRP) {
                             excl_tiles += gg4[q] & ((1ULL << 42) - 1);
                             g--;
+                            continue;
+                        }
+                        stall = true;
+                        break;
+                    }
+                    if (stall) nstall++;
+                    niter++;
+                }
+            }
+            /* 3) whole supers below: one load consumes OSW_SUP ranked
+             * tiles; the super's last-tile descriptor is the INC shortcut
+             * that ends the walk within the deferral window */
+            if (!done && vb >= OSW_SUP) {
+                gdesc_t *scol =
+                    (gdesc_t *)(ff + (uint64_t)((nblocks + OSW_GRP - 1) >> OSW_GRP_LG) * 256 + t);
+                int64_t su = (int64_t)(vb >> OSW_SUP_LG) - 1;
+                while (su >= 0) {
+                    if (++spins > (1u << 26)) { *d_abort = 1; break; }
+                    int sa = (su >= 3) ? 4 : (int)(su + 1);
+                    unsigned long long s0, s1 = 0, s2 = 0, s3 = 0;
+                    s0 = __hip_atomic_load(scol + (uint64_t)su * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    if (sa > 1) s1 = __hip_atomic_load(scol + (uint64_t)(su - 1) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    if (sa > 2) s2 = __hip_atomic_load(scol + (uint64_t)(su - 2) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    if (sa > 3) s3 = __hip_atomic_load(scol + (uint64_t)(su - 3) * 256, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    unsigned long long se = __hip_atomic_load(
+                        col + (((uint64_t)su + 1) * OSW_SUP - 1) * 256, __ATOMIC_RELAXED,
+                        __HIP_MEMORY_SCOPE_AGENT);
+                    if ((se >> 62) == 2 && OSW_TAG(se) == (uint32_t)(ptag & 0x3F)) {
+                        excl_tiles += se & OSW_CNT_MASK; /* covers [0,(su+1)*SUP-1] */
+                        break;
+                    }
+                    unsigned long long ss4[4] = {s0, s1, s2, s3};
+                    bool stall = false;
+                    for (int q = 0; q < sa; ++q) {
+                        if ((ss4[q] >> 42) == OSW_SUP) {
+                            excl_tiles += ss4[q] & ((1ULL << 42) - 1);
+                            su--;
                             continue;
                         }
                         stall = true;
@@ -848,7 +897,9 @@ static hipError_t scatter_pass_osw(hipStream_t s, const uint64_t *in_k, const ui
     if (phase_prof_mode() == 2) ptag |= 0x100; /* enable walk counters */
     uint32_t nb = nblocks_for(n);
     /* desc is zeroed once per SORT call (pass tags invalidate stale words) */
-    HIP_TRY(hipMemsetAsync(ff, 0, (size_t)((nb + OSW_GRP - 1) / OSW_GRP) * 256 * 8, s));
+    HIP_TRY(hipMemsetAsync(ff, 0,
+                           ((size_t)((nb + OSW_GRP - 1) / OSW_GRP) +
+                            (size_t)((nb + OSW_SUP - 1) / OSW_SUP)) * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(ticket, 0, 4, s));
     ProfScope ps(prof_name, s);
     size_t sh = (has_vals ? 16 : 8) * (size_t)TILE +
@@ -951,8 +1002,9 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *ticket = (uint32_t *)ws.take(256);
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_abort = (int *)ws.take(256);
-    unsigned long long *ff_d =
-        (unsigned long long *)ws.take((size_t)((nb + OSW_GRP - 1) / OSW_GRP) * 256 * 8);
+    unsigned long long *ff_d = (unsigned long long *)ws.take(
+        ((size_t)((nb + OSW_GRP - 1) / OSW_GRP) +
+         (size_t)((nb + OSW_SUP - 1) / OSW_SUP)) * 256 * 8);
     if (!pA || !pB || !desc || !gbase_d || !ticket || !h8 || !d_abort || !ff_d)
         return hipErrorOutOfMemory;
     HIP_TRY(hipMemsetAsync(d_abort, 0, 4, s));
@@ -1193,8 +1245,9 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     uint32_t *h8 = (uint32_t *)ws.take(8 * 256 * 4);
     int *d_err = (int *)ws.take(256);
     int *d_abort = (int *)ws.take(256);
-    unsigned long long *ff_d =
-        (unsigned long long *)ws.take((size_t)((nb + OSW_GRP - 1) / OSW_GRP) * 256 * 8);
+    unsigned long long *ff_d = (unsigned long long *)ws.take(
+        ((size_t)((nb + OSW_GRP - 1) / OSW_GRP) +
+         (size_t)((nb + OSW_SUP - 1) / OSW_SUP)) * 256 * 8);
     uint32_t *h32buf = (uint32_t *)ws.take(n * 4);
     unsigned long long *wl = (unsigned long long *)ws.take(CLEANUP_WL_CAP * 8);
     uint32_t *wl_count = (uint32_t *)ws.take(256);
@@ -1914,7 +1967,7 @@ size_t ws_bytes_for(uint64_t n) {
     /* scan recursion partials: nb/TILE + nb/TILE^2 + ... < nb/2048 */
     b += (((size_t)nb / 2048 + 4096) * 4 + 255) & ~255ULL;
     b += ((size_t)257 * 4 + 255) & ~255ULL;       /* partition starts */
-    b += ((size_t)((nb + 15) / 16) * 2048 + 255) & ~255ULL; /* group descriptors */
+    b += ((size_t)((nb + 15) / 16 + nb / 256 + 2) * 2048 + 255) & ~255ULL; /* group+super descriptors */
     b += (size_t)CLEANUP_WL_CAP * 8 + 512;        /* cleanup long-run worklist */
     b += ((size_t)nb * BLOCK * 12 + 255) & ~255ULL; /* f64 lead partials (seg, OP 2) */
     b += 1 << 20;                                 /* slack */
