@@ -715,6 +715,7 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
                         __HIP_MEMORY_SCOPE_AGENT);
                     if ((ge >> 62) == 2 && OSW_TAG(ge) == (uint32_t)(ptag & 0x3F)) {
                         excl_tiles += ge & OSW_CNT_MASK; /* covers [0, (g+1)*GRP-1] */
+                        done = true; /* WHOLE prefix consumed: supers must not re-add it */
                         break;
                     }
                     unsigned long long gg4[4] = {g0, g1, g2, g3};
