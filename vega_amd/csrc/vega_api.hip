@@ -663,7 +663,7 @@ int vega_gpu_group_by_key(vega_ctx_t *c, vega_rdd_t rdd, uint32_t nparts,
     Ws ws(c->ws, c->ws_bytes);
     const uint64_t *sk, *sv;
     CTX_TRY(c, group_sort_u64(c->stream, (const uint64_t *)r->d_k, (const uint64_t *)r->d_v,
-                              r->n, 0, nullptr, ws, &sk, &sv));
+                              r->n, 0, nullptr, 0, nullptr, ws, &sk, &sv));
     /* values in grouped order */
     if (r->n)
         CTX_TRY(c, hipMemcpyAsync(o->d_v2, sv, r->n * 8, hipMemcpyDeviceToDevice, c->stream));

@@ -59,15 +59,19 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
  * adaptive skipped key sort vs 40-bit hash sort + collision-run cleanup */
 /* order_tag out: 0 = full-key unsigned order, 4 = (h32,key) lex order.
  * force_hbytes: 0 adaptive (reduce), 4 pinned order (joins). */
+/* want_packed: hash path keeps the final pass's interleaved (k,v) layout
+ * (cheaper writeout); *out_packed reports whether the result IS packed
+ * (key i at res_k[2i]) — the narrow-key and fallback paths stay SoA. */
 hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
-                          uint64_t n, int force_hbytes, int *order_tag, Ws &ws,
+                          uint64_t n, int force_hbytes, int *order_tag,
+                          int want_packed, int *out_packed, Ws &ws,
                           const uint64_t **res_k, const uint64_t **res_v);
 
 /* segmented aggregate over key-sorted rows: one output row per equal-key
  * run. op: VEGA_OP_*. Returns #segments in *h_nout (after stream sync). */
 hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t n,
                       int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws,
-                      bool v_prezeroed = false);
+                      bool v_prezeroed = false, bool packed = false);
 
 /* grouping sort + segmented aggregate, accumulator init overlapped on a
  * side stream (the fast path reduce_by_key / group_count use) */

@@ -814,6 +814,8 @@ __global__ __launch_bounds__(512) void k_scatter_osw(
              * is load-bearing (nontemporal regressed 33%) */
             if (OUT_PK) {
                 ((ulonglong2 *)out_k)[gpos] = make_ulonglong2(k, v);
+                if (h32_out) /* packed final pass still feeds the cleanup */
+                    h32_out[gpos] = (uint32_t)vega_hash_u64(k);
             } else {
                 out_k[gpos] = k;
                 if (HAS_VALS) out_v[gpos] = v;
@@ -1107,8 +1109,11 @@ hipError_t radix_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 #define CLEANUP_WALK_CAP 1024
 #define CLEANUP_WL_CAP 65536
 __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
-                                uint64_t n, int strict, int *err,
+                                uint64_t n, int packed, int strict, int *err,
                                 unsigned long long *wl, uint32_t *wl_count) {
+    /* packed: rows are interleaved (k,v); key x at k[ST x], value beside it */
+    const int ST = packed ? 2 : 1;
+    uint64_t *vb = packed ? k + 1 : v;
     uint64_t nchunks = (n + 3) / 4;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
@@ -1146,15 +1151,15 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
             if (!strict && len == 2) continue; /* nothing to group either way */
             bool fix;
             if (strict) {
-                uint64_t k0 = k[gi];
+                uint64_t k0 = k[ST * gi];
                 bool dirty = false;
-                for (uint64_t x = gi + 1; x < je && !dirty; x++) dirty = (k[x] != k0);
+                for (uint64_t x = gi + 1; x < je && !dirty; x++) dirty = (k[ST * x] != k0);
                 fix = dirty;
             } else {
                 uint32_t breaks = 0;
-                uint64_t prev = k[gi];
+                uint64_t prev = k[ST * gi];
                 for (uint64_t x = gi + 1; x < je && breaks < 2; x++) {
-                    uint64_t cx = k[x];
+                    uint64_t cx = k[ST * x];
                     breaks += cx != prev;
                     prev = cx;
                 }
@@ -1163,15 +1168,15 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
             if (!fix) continue;
             if (len > 64) { *err = 1; continue; }
             for (uint64_t x = gi + 1; x < je; x++) {
-                uint64_t kx = k[x], vx = v[x];
+                uint64_t kx = k[ST * x], vx = vb[ST * x];
                 uint64_t y = x;
-                while (y > gi && k[y - 1] > kx) {
-                    k[y] = k[y - 1];
-                    v[y] = v[y - 1];
+                while (y > gi && k[ST * (y - 1)] > kx) {
+                    k[ST * y] = k[ST * (y - 1)];
+                    vb[ST * y] = vb[ST * (y - 1)];
                     y--;
                 }
-                k[y] = kx;
-                v[y] = vx;
+                k[ST * y] = kx;
+                vb[ST * y] = vx;
             }
         }
     }
@@ -1184,9 +1189,10 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
  * *err and the caller falls back to the full key sort. The common long run
  * (a hot key, all rows equal) passes with zero breaks. */
 __global__ void k_group_cleanup_long(const uint64_t *k, const uint32_t *h32,
-                                     uint64_t n, int strict, int *err,
+                                     uint64_t n, int packed, int strict, int *err,
                                      const unsigned long long *wl,
                                      const uint32_t *wl_count) {
+    const int ST = packed ? 2 : 1;
     __shared__ unsigned long long s_end;
     __shared__ unsigned int s_breaks;
     uint32_t cnt = *wl_count;
@@ -1202,11 +1208,11 @@ __global__ void k_group_cleanup_long(const uint64_t *k, const uint32_t *h32,
             __syncthreads();
         }
         uint64_t je = s_end;
-        uint64_t k0 = k[gi];
+        uint64_t k0 = k[ST * gi];
         for (uint64_t base = gi + 1; base < je && s_breaks < 2; base += blockDim.x) {
             uint64_t x = base + threadIdx.x;
             if (x < je) {
-                bool br = strict ? (k[x] != k0) : (k[x] != k[x - 1]);
+                bool br = strict ? (k[ST * x] != k0) : (k[ST * x] != k[ST * (x - 1)]);
                 if (br) atomicAdd(&s_breaks, 1u);
             }
             __syncthreads();
@@ -1229,10 +1235,12 @@ __global__ void k_group_cleanup_long(const uint64_t *k, const uint32_t *h32,
  * pins the hash order regardless of n (joins need a stable comparator);
  * 0 = adaptive (reduce path). */
 hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *in_v,
-                          uint64_t n, int force_hbytes, int *order_tag, Ws &ws,
+                          uint64_t n, int force_hbytes, int *order_tag,
+                          int want_packed, int *out_packed, Ws &ws,
                           const uint64_t **res_k, const uint64_t **res_v) {
     *res_k = in_k;
     *res_v = in_v;
+    if (out_packed) *out_packed = 0;
     if (order_tag) *order_tag = 0;
     if (n <= 1) return hipSuccess;
     if (n >= (1ULL << 32)) return hipErrorNotSupported; /* u32 hist/scan limit (vega_gpu.h) */
@@ -1348,15 +1356,18 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
     const int hbytes = force_hbytes ? force_hbytes : ((n <= (1ULL << 30)) ? 4 : 5);
     int active5 = 0;
     HIP_TRY(exact_hists(true, hbytes, &active5));
+    const bool keep_pk = want_packed != 0; /* final pass stays packed:
+        the SoA+h32 unpack pass measured 12.1 ms vs 8.6 interior */
     for (int i = 0; i < hbytes; ++i) {
-        bool in_pk = i > 0, out_pk = i < hbytes - 1;
+        bool in_pk = i > 0, out_pk = keep_pk || (i < hbytes - 1);
+        bool last = i == hbytes - 1;
         uint64_t *dbuf = ((i & 1) == 0) ? pA : pB;
         uint64_t *dk = dbuf;
         uint64_t *dv = out_pk ? nullptr : dbuf + n;
         HashByteDigit df{8 * i};
         HIP_TRY(scatter_pass_osw(s, i == 0 ? in_k : cur_k, i == 0 ? in_v : nullptr,
                                  n, gbase_d + i * 256, desc, ticket,
-                                 dk, dv, out_pk ? nullptr : h32buf,
+                                 dk, dv, last ? h32buf : nullptr,
                                  d_abort, ff_d, ptag_ctr++, true, in_pk, out_pk, df, "radix_scatter"));
         cur_k = dk;
         cur_v = out_pk ? nullptr : dk + n;
@@ -1367,11 +1378,11 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         ProfScope ps("group_cleanup", s);
         uint32_t gb = nb < 2048 ? nb : 2048;
         hipLaunchKernelGGL(k_group_cleanup, dim3(gb), dim3(BLOCK), 0, s,
-                           (uint64_t *)cur_k, (uint64_t *)cur_v, h32buf, n, strict,
-                           d_err, wl, wl_count);
+                           (uint64_t *)cur_k, (uint64_t *)cur_v, h32buf, n,
+                           keep_pk ? 1 : 0, strict, d_err, wl, wl_count);
         HIP_TRY(hipGetLastError());
         hipLaunchKernelGGL(k_group_cleanup_long, dim3(512), dim3(BLOCK), 0, s,
-                           cur_k, h32buf, n, strict, d_err, wl, wl_count);
+                           cur_k, h32buf, n, keep_pk ? 1 : 0, strict, d_err, wl, wl_count);
         HIP_TRY(hipGetLastError());
     }
     /* one readback for both flags: abort is final once the scatter passes
@@ -1390,6 +1401,8 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
         HIP_TRY(hipMemcpyAsync(&ab2, d_abort, 4, hipMemcpyDeviceToHost, s));
         HIP_TRY(hipStreamSynchronize(s));
         if (ab2) return hipErrorUnknown;
+    } else if (keep_pk && out_packed) {
+        *out_packed = 1;
     }
     if (order_tag && !err) *order_tag = (hbytes == 4) ? 4 : 0;
     *res_k = cur_k;
@@ -1400,6 +1413,9 @@ hipError_t group_sort_u64(hipStream_t s, const uint64_t *in_k, const uint64_t *i
 /* ------------------------------------------------------------------ */
 /* segmented reduce over key-sorted rows                               */
 
+/* PK: rows are interleaved 16-B (k,v) pairs — key i at k[2i] (lets the
+ * final hash pass stay packed; the SoA unpack pass cost 12 vs 8.6 ms) */
+template <bool PK>
 __global__ void k_head_count(const uint64_t *k, uint64_t n, uint32_t *hc) {
     __shared__ uint32_t wsum[BLOCK / 64];
     uint64_t tbase = (uint64_t)blockIdx.x * TILE;
@@ -1407,7 +1423,8 @@ __global__ void k_head_count(const uint64_t *k, uint64_t n, uint32_t *hc) {
 #pragma unroll
     for (int j = 0; j < IPT; ++j) {
         uint64_t idx = tbase + (uint64_t)j * BLOCK + threadIdx.x;
-        if (idx < n) c += (idx == 0) || (k[idx] != k[idx - 1]);
+        if (idx < n)
+            c += (idx == 0) || (k[PK ? 2 * idx : idx] != k[PK ? 2 * (idx - 1) : idx - 1]);
     }
     for (int off = 32; off > 0; off >>= 1) c += __shfl_down(c, off);
     int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
@@ -1432,7 +1449,7 @@ __global__ void k_head_count(const uint64_t *k, uint64_t n, uint32_t *hc) {
  * order — a fixed summation shape for a given n, so results are bit-stable
  * run to run (and within 1e-6 rel of the reference's sequential merge,
  * pair_rdd.rs:74-78). */
-template <int OP>
+template <int OP, bool PK>
 __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     const uint64_t *__restrict__ k, const void *__restrict__ vv, uint64_t n,
     const uint32_t *__restrict__ head_base, int64_t *__restrict__ out_k,
@@ -1448,28 +1465,38 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
 
     uint64_t kk[IPT], sv[IPT];
     if (c0g + IPT <= n) { /* interior chunk: unguarded 16-B vector loads */
-        const ulonglong2 *kp = reinterpret_cast<const ulonglong2 *>(k + c0g);
-        const ulonglong2 *vp = reinterpret_cast<const ulonglong2 *>(v + c0g);
+        if (PK) {
+            const ulonglong2 *rp = reinterpret_cast<const ulonglong2 *>(k) + c0g;
 #pragma unroll
-        for (int j2 = 0; j2 < IPT / 2; ++j2) {
-            ulonglong2 t2 = kp[j2];
-            kk[2 * j2] = t2.x;
-            kk[2 * j2 + 1] = t2.y;
-            if (NEED_V) {
-                ulonglong2 u2 = vp[j2];
-                sv[2 * j2] = u2.x;
-                sv[2 * j2 + 1] = u2.y;
+            for (int j = 0; j < IPT; ++j) {
+                ulonglong2 r2 = rp[j];
+                kk[j] = r2.x;
+                sv[j] = r2.y;
+            }
+        } else {
+            const ulonglong2 *kp = reinterpret_cast<const ulonglong2 *>(k + c0g);
+            const ulonglong2 *vp = reinterpret_cast<const ulonglong2 *>(v + c0g);
+#pragma unroll
+            for (int j2 = 0; j2 < IPT / 2; ++j2) {
+                ulonglong2 t2 = kp[j2];
+                kk[2 * j2] = t2.x;
+                kk[2 * j2 + 1] = t2.y;
+                if (NEED_V) {
+                    ulonglong2 u2 = vp[j2];
+                    sv[2 * j2] = u2.x;
+                    sv[2 * j2 + 1] = u2.y;
+                }
             }
         }
     } else {
 #pragma unroll
         for (int j = 0; j < IPT; ++j) {
             uint64_t gi = c0g + j;
-            kk[j] = (gi < n) ? k[gi] : ~0ULL;
-            if (NEED_V) sv[j] = (gi < n) ? v[gi] : 0;
+            kk[j] = (gi < n) ? k[PK ? 2 * gi : gi] : ~0ULL;
+            if (NEED_V || PK) sv[j] = (gi < n) ? (PK ? k[2 * gi + 1] : v[gi]) : 0;
         }
     }
-    uint64_t prev = (c0g > 0 && c0g <= n) ? k[c0g - 1] : 0;
+    uint64_t prev = (c0g > 0 && c0g <= n) ? k[PK ? 2 * (c0g - 1) : c0g - 1] : 0;
 
     /* count heads in the chunk */
     uint32_t cnt = 0;
@@ -1539,7 +1566,7 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
      * length-1 runs with pure vector stores — no branches, no atomics. */
     bool fast = false;
     if (cnt == IPT && c0g + IPT <= n) {
-        uint64_t nk = (c0g + IPT < n) ? k[c0g + IPT] : ~kk[IPT - 1];
+        uint64_t nk = (c0g + IPT < n) ? k[PK ? 2 * (c0g + IPT) : c0g + IPT] : ~kk[IPT - 1];
         fast = (nk != kk[IPT - 1]);
     }
     int64_t segid = (int64_t)head_base[blockIdx.x] + excl - 1;
@@ -1651,7 +1678,7 @@ __global__ void k_f64_seg_combine(const uint32_t *lead_seg, const double *lead_p
 
 hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t n,
                       int op, uint64_t *out_k, void *out_v, uint64_t *h_nout, Ws &ws,
-                      bool v_prezeroed) {
+                      bool v_prezeroed, bool packed) {
     if (n == 0) { *h_nout = 0; return hipSuccess; }
     if (n >= (1ULL << 32)) return hipErrorNotSupported; /* u32 head scan limit */
     uint32_t nb = nblocks_for(n);
@@ -1667,7 +1694,10 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     }
     {
         ProfScope ps("head_count", s);
-        hipLaunchKernelGGL(k_head_count, dim3(nb), dim3(BLOCK), 0, s, k, n, hc);
+        if (packed)
+            hipLaunchKernelGGL((k_head_count<true>), dim3(nb), dim3(BLOCK), 0, s, k, n, hc);
+        else
+            hipLaunchKernelGGL((k_head_count<false>), dim3(nb), dim3(BLOCK), 0, s, k, n, hc);
         HIP_TRY(hipGetLastError());
     }
     HIP_TRY(hipMemsetAsync(hc + nb, 0, 4, s));
@@ -1694,14 +1724,21 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     {
         ProfScope ps("seg_emit", s);
         size_t sh = 0;
-        switch (op) {
-        case 0: hipLaunchKernelGGL(k_seg_emit<0>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
-        case 1: hipLaunchKernelGGL(k_seg_emit<1>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
-        case 2: hipLaunchKernelGGL(k_seg_emit<2>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
-        case 3: hipLaunchKernelGGL(k_seg_emit<3>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
-        case 4: hipLaunchKernelGGL(k_seg_emit<4>, dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part); break;
+#define SEG_LAUNCH(OPN, PKB) hipLaunchKernelGGL((k_seg_emit<OPN, PKB>), dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part)
+        switch (op * 2 + (packed ? 1 : 0)) {
+        case 0: SEG_LAUNCH(0, false); break;
+        case 1: SEG_LAUNCH(0, true); break;
+        case 2: SEG_LAUNCH(1, false); break;
+        case 3: SEG_LAUNCH(1, true); break;
+        case 4: SEG_LAUNCH(2, false); break;
+        case 5: SEG_LAUNCH(2, true); break;
+        case 6: SEG_LAUNCH(3, false); break;
+        case 7: SEG_LAUNCH(3, true); break;
+        case 8: SEG_LAUNCH(4, false); break;
+        case 9: SEG_LAUNCH(4, true); break;
         default: return hipErrorInvalidValue;
         }
+#undef SEG_LAUNCH
         HIP_TRY(hipGetLastError());
     }
     if (op == 2) {
@@ -1724,9 +1761,10 @@ hipError_t group_sort_reduce(hipStream_t s, const uint64_t *in_k, const uint64_t
                              uint64_t *h_nout, Ws &ws) {
     if (n == 0) { *h_nout = 0; return hipSuccess; }
     const uint64_t *sk, *sv;
-    hipError_t e = group_sort_u64(s, in_k, in_v, n, 0, nullptr, ws, &sk, &sv);
+    int pk = 0;
+    hipError_t e = group_sort_u64(s, in_k, in_v, n, 0, nullptr, 1, &pk, ws, &sk, &sv);
     if (e != hipSuccess) return e;
-    return seg_reduce(s, sk, sv, n, op, out_k, out_v, h_nout, ws, false);
+    return seg_reduce(s, sk, sv, n, op, out_k, out_v, h_nout, ws, false, pk != 0);
 }
 
 /* ------------------------------------------------------------------ */
@@ -1948,7 +1986,7 @@ hipError_t group_pairs_inplace(hipStream_t s, int64_t *keys, int64_t *vals,
                                uint64_t n, int *order_tag, Ws &ws) {
     const uint64_t *rk, *rv;
     HIP_TRY(group_sort_u64(s, (const uint64_t *)keys, (const uint64_t *)vals, n,
-                           /*force_hbytes=*/4, order_tag, ws, &rk, &rv));
+                           /*force_hbytes=*/4, order_tag, 0, nullptr, ws, &rk, &rv));
     if ((const uint64_t *)keys != rk) {
         HIP_TRY(hipMemcpyAsync(keys, rk, n * 8, hipMemcpyDeviceToDevice, s));
         HIP_TRY(hipMemcpyAsync(vals, rv, n * 8, hipMemcpyDeviceToDevice, s));
