@@ -1114,19 +1114,21 @@ __global__ void k_group_cleanup(uint64_t *k, uint64_t *v, const uint32_t *h32,
     /* packed: rows are interleaved (k,v); key x at k[ST x], value beside it */
     const int ST = packed ? 2 : 1;
     uint64_t *vb = packed ? k + 1 : v;
-    uint64_t nchunks = (n + 7) / 8;
+    uint64_t nchunks = (n + 15) / 16;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    /* 8-row chunks = two independent uint4 loads in flight per iteration:
+    /* 16-row chunks = four independent uint4 loads in flight per iteration:
      * the scan was latency-bound at one (640 GB/s on a 4 GB sweep) */
     for (uint64_t c = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c < nchunks; c += stride) {
-        uint64_t i0 = 8 * c;
-        int m = (int)((n - i0 < 8) ? (n - i0) : 8);
-        uint32_t hh[10];
-        if (m == 8) {
-            uint4 hv = ((const uint4 *)h32)[2 * c];
-            uint4 hw = ((const uint4 *)h32)[2 * c + 1];
-            hh[1] = hv.x; hh[2] = hv.y; hh[3] = hv.z; hh[4] = hv.w;
-            hh[5] = hw.x; hh[6] = hw.y; hh[7] = hw.z; hh[8] = hw.w;
+        uint64_t i0 = 16 * c;
+        int m = (int)((n - i0 < 16) ? (n - i0) : 16);
+        uint32_t hh[18];
+        if (m == 16) {
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                uint4 hv = ((const uint4 *)h32)[4 * c + q];
+                hh[4 * q + 1] = hv.x; hh[4 * q + 2] = hv.y;
+                hh[4 * q + 3] = hv.z; hh[4 * q + 4] = hv.w;
+            }
         } else {
             for (int j = 0; j < m; ++j) hh[j + 1] = h32[i0 + j];
         }
