@@ -1455,26 +1455,30 @@ __global__ void k_head_count(const uint64_t *k, uint64_t n, uint32_t *hc) {
  * order — a fixed summation shape for a given n, so results are bit-stable
  * run to run (and within 1e-6 rel of the reference's sequential merge,
  * pair_rdd.rs:74-78). */
+/* 512 threads x 8-row chunks per 4096-row tile: shorter per-thread load
+ * chains than 256x16 (the chunk loads are the latency chain) */
+#define SEG_B 512
+#define SEG_IPT 8
 template <int OP, bool PK>
-__global__ __launch_bounds__(BLOCK) void k_seg_emit(
+__global__ __launch_bounds__(SEG_B) void k_seg_emit(
     const uint64_t *__restrict__ k, const void *__restrict__ vv, uint64_t n,
     const uint32_t *__restrict__ head_base, int64_t *__restrict__ out_k,
     void *__restrict__ out_vv, uint32_t *__restrict__ lead_seg,
     double *__restrict__ lead_part) {
-    __shared__ uint32_t wsc[BLOCK / 64];
+    __shared__ uint32_t wsc[SEG_B / 64];
     constexpr bool NEED_V = (OP != 1);
 
     const int t = threadIdx.x, lane = t & 63, w = t >> 6;
     const uint64_t tbase = (uint64_t)blockIdx.x * TILE;
     const uint64_t *v = (const uint64_t *)vv;
-    const uint64_t c0g = tbase + (uint64_t)t * IPT; /* chunk start, global */
+    const uint64_t c0g = tbase + (uint64_t)t * SEG_IPT; /* chunk start, global */
 
-    uint64_t kk[IPT], sv[IPT];
-    if (c0g + IPT <= n) { /* interior chunk: unguarded 16-B vector loads */
+    uint64_t kk[SEG_IPT], sv[SEG_IPT];
+    if (c0g + SEG_IPT <= n) { /* interior chunk: unguarded 16-B vector loads */
         if (PK) {
             const ulonglong2 *rp = reinterpret_cast<const ulonglong2 *>(k) + c0g;
 #pragma unroll
-            for (int j = 0; j < IPT; ++j) {
+            for (int j = 0; j < SEG_IPT; ++j) {
                 ulonglong2 r2 = rp[j];
                 kk[j] = r2.x;
                 sv[j] = r2.y;
@@ -1483,7 +1487,7 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
             const ulonglong2 *kp = reinterpret_cast<const ulonglong2 *>(k + c0g);
             const ulonglong2 *vp = reinterpret_cast<const ulonglong2 *>(v + c0g);
 #pragma unroll
-            for (int j2 = 0; j2 < IPT / 2; ++j2) {
+            for (int j2 = 0; j2 < SEG_IPT / 2; ++j2) {
                 ulonglong2 t2 = kp[j2];
                 kk[2 * j2] = t2.x;
                 kk[2 * j2 + 1] = t2.y;
@@ -1496,7 +1500,7 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
         }
     } else {
 #pragma unroll
-        for (int j = 0; j < IPT; ++j) {
+        for (int j = 0; j < SEG_IPT; ++j) {
             uint64_t gi = c0g + j;
             kk[j] = (gi < n) ? k[PK ? 2 * gi : gi] : ~0ULL;
             if (NEED_V || PK) sv[j] = (gi < n) ? (PK ? k[2 * gi + 1] : v[gi]) : 0;
@@ -1507,7 +1511,7 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     /* count heads in the chunk */
     uint32_t cnt = 0;
 #pragma unroll
-    for (int j = 0; j < IPT; ++j) {
+    for (int j = 0; j < SEG_IPT; ++j) {
         uint64_t gi = c0g + j;
         if (gi < n) {
             uint64_t pk = (j > 0) ? kk[j - 1] : prev;
@@ -1526,13 +1530,13 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
     for (int i = 0; i < w; ++i) excl += wsc[i];
 
     if (OP == 2) { /* deterministic f64 path (see kernel comment) */
-        uint64_t chunk_id = (uint64_t)blockIdx.x * BLOCK + t;
+        uint64_t chunk_id = (uint64_t)blockIdx.x * SEG_B + t;
         int64_t sid = (int64_t)head_base[blockIdx.x] + excl - 1; /* run open at chunk start */
         if (c0g >= n) {
             lead_seg[chunk_id] = 0xFFFFFFFFu;
             return;
         }
-        int nvalid = (int)((n - c0g < IPT) ? (n - c0g) : IPT);
+        int nvalid = (int)((n - c0g < SEG_IPT) ? (n - c0g) : SEG_IPT);
         int fh = -1;
         for (int j = 0; j < nvalid; ++j) {
             uint64_t pk2 = (j > 0) ? kk[j - 1] : prev;
@@ -1571,9 +1575,9 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
      * continue into the next chunk (the dominant C1 shape) emits its 16
      * length-1 runs with pure vector stores — no branches, no atomics. */
     bool fast = false;
-    if (cnt == IPT && c0g + IPT <= n) {
-        uint64_t nk = (c0g + IPT < n) ? k[PK ? 2 * (c0g + IPT) : c0g + IPT] : ~kk[IPT - 1];
-        fast = (nk != kk[IPT - 1]);
+    if (cnt == SEG_IPT && c0g + SEG_IPT <= n) {
+        uint64_t nk = (c0g + SEG_IPT < n) ? k[PK ? 2 * (c0g + SEG_IPT) : c0g + SEG_IPT] : ~kk[SEG_IPT - 1];
+        fast = (nk != kk[SEG_IPT - 1]);
     }
     int64_t segid = (int64_t)head_base[blockIdx.x] + excl - 1;
     int64_t acc_i = (OP == 3) ? INT64_MAX : (OP == 4) ? INT64_MIN : 0;
@@ -1587,14 +1591,14 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
             ulonglong2 *ok2 = (ulonglong2 *)okp;
             ulonglong2 *ov2 = (ulonglong2 *)ovp;
 #pragma unroll
-            for (int j = 0; j < IPT / 2; ++j) {
+            for (int j = 0; j < SEG_IPT / 2; ++j) {
                 ok2[j] = make_ulonglong2(kk[2 * j], kk[2 * j + 1]);
                 if (OP == 1) ov2[j] = make_ulonglong2(1, 1);
                 else ov2[j] = make_ulonglong2(sv[2 * j], sv[2 * j + 1]);
             }
         } else {
 #pragma unroll
-            for (int j = 0; j < IPT; ++j) {
+            for (int j = 0; j < SEG_IPT; ++j) {
                 okp[j] = (int64_t)kk[j];
                 ovp[j] = (OP == 1) ? 1ULL : sv[j];
             }
@@ -1603,7 +1607,7 @@ __global__ __launch_bounds__(BLOCK) void k_seg_emit(
          * lane but still participates in the shfl lanes */
     } else
 #pragma unroll
-    for (int j = 0; j < IPT; ++j) {
+    for (int j = 0; j < SEG_IPT; ++j) {
         uint64_t gi = c0g + j;
         if (gi >= n) break;
         uint64_t key = kk[j];
@@ -1690,7 +1694,7 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     uint32_t nb = nblocks_for(n);
     uint32_t *hc = (uint32_t *)ws.take(((size_t)nb + 1) * 4);
     if (!hc) return hipErrorOutOfMemory;
-    uint64_t nchunks = (uint64_t)nb * BLOCK;
+    uint64_t nchunks = (uint64_t)nb * 512; /* SEG_B chunks per tile */
     uint32_t *lead_seg = nullptr;
     double *lead_part = nullptr;
     if (op == 2) {
@@ -1730,7 +1734,7 @@ hipError_t seg_reduce(hipStream_t s, const uint64_t *k, const void *v, uint64_t 
     {
         ProfScope ps("seg_emit", s);
         size_t sh = 0;
-#define SEG_LAUNCH(OPN, PKB) hipLaunchKernelGGL((k_seg_emit<OPN, PKB>), dim3(nb), dim3(BLOCK), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part)
+#define SEG_LAUNCH(OPN, PKB) hipLaunchKernelGGL((k_seg_emit<OPN, PKB>), dim3(nb), dim3(512), sh, s, k, v, n, hc, (int64_t *)out_k, out_v, lead_seg, lead_part)
         switch (op * 2 + (packed ? 1 : 0)) {
         case 0: SEG_LAUNCH(0, false); break;
         case 1: SEG_LAUNCH(0, true); break;
@@ -2014,7 +2018,7 @@ size_t ws_bytes_for(uint64_t n) {
     b += ((size_t)257 * 4 + 255) & ~255ULL;       /* partition starts */
     b += ((size_t)((nb + 15) / 16 + nb / 256 + 2) * 2048 + 255) & ~255ULL; /* group+super descriptors */
     b += (size_t)CLEANUP_WL_CAP * 8 + 512;        /* cleanup long-run worklist */
-    b += ((size_t)nb * BLOCK * 12 + 255) & ~255ULL; /* f64 lead partials (seg, OP 2) */
+    b += ((size_t)nb * 512 * 12 + 255) & ~255ULL; /* f64 lead partials (seg, OP 2) */
     b += 1 << 20;                                 /* slack */
     return b;
 }
