@@ -52,3 +52,30 @@ def test_sort_random_cfg(ctx, n, bits, seed):
     order = np.argsort(k, kind="stable")
     assert (sk == k[order]).all() and (sv == v[order]).all()
     rdd.free(); srt.free()
+
+
+def test_empty_inputs(ctx):
+    """n == 0 through every handle-API op (the reference's tests cover empty
+    partitions; ragged/empty inputs must not fault or mis-count)"""
+    from vega_amd import gpu
+    e = np.empty(0, dtype=np.int64)
+    rdd = ctx.make_rdd(e, e)
+    red = rdd.reduce_by_key(gpu.OP_SUM_I64)
+    assert red.count() == 0
+    k, v = red.collect()
+    assert len(k) == 0
+    srt = rdd.sort_by_key()
+    assert srt.count() == 0
+    gk, off, gv = rdd.group_by_key()
+    assert len(gk) == 0 and len(gv) == 0
+    other = ctx.make_rdd(np.array([1, 2], dtype=np.int64),
+                         np.array([3, 4], dtype=np.int64))
+    j = rdd.join(other)
+    assert j.count() == 0
+    inter = rdd.intersection(other)
+    assert inter.count() == 0
+    sub = other.subtract(rdd)
+    sk, _ = sub.collect()
+    assert sorted(sk.tolist()) == [1, 2]
+    for r in (rdd, red, srt, other, j, inter, sub):
+        r.free()
