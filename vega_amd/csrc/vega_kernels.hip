@@ -2263,56 +2263,23 @@ __global__ void k_join_count(const int64_t *ak, uint64_t na, const int64_t *bk,
         __syncthreads();
         const uint64_t wlo = s_lo, whi = s_hi;
         __syncthreads(); /* s_lo/s_hi free for the next span */
-        /* each thread owns 16 CONSECUTIVE (sorted) A rows and advances a B
-         * cursor merge-style: linear steps while the gap is small, binary
-         * search when it is not — sequential loads instead of 24 dependent
-         * cold probes per row */
-        for (uint64_t i0 = a0 + (uint64_t)threadIdx.x * 16; i0 < a1;
-             i0 += (uint64_t)blockDim.x * 16) {
-            uint64_t iend = (a1 - i0 < 16) ? a1 : i0 + 16;
-            /* seed the cursor: binary lower bound of the first key */
-            int64_t k0 = ak[i0];
+        for (uint64_t i = a0 + threadIdx.x; i < a1; i += blockDim.x) {
+            int64_t k = ak[i];
+            /* lower bound within the span window */
             uint64_t lo = wlo, hi = whi;
             while (lo < hi) {
                 uint64_t m = (lo + hi) >> 1;
-                if (join_less(mode, bk[m], k0)) lo = m + 1; else hi = m;
+                if (join_less(mode, bk[m], k)) lo = m + 1; else hi = m;
             }
-            for (uint64_t i = i0; i < iend; ++i) {
-                int64_t k = ak[i];
-                /* lower bound: linear advance (bounded), binary fallback */
-                int steps = 0;
-                while (lo < whi && join_less(mode, bk[lo], k)) {
-                    ++lo;
-                    if (++steps == 32) {
-                        uint64_t l2 = lo, h2 = whi;
-                        while (l2 < h2) {
-                            uint64_t m = (l2 + h2) >> 1;
-                            if (join_less(mode, bk[m], k)) l2 = m + 1; else h2 = m;
-                        }
-                        lo = l2;
-                        break;
-                    }
-                }
-                uint64_t lb = lo;
-                /* upper bound: a few linear steps, then binary for long runs */
-                uint64_t ub = lb;
-                steps = 0;
-                while (ub < whi && !join_less(mode, k, bk[ub])) {
-                    ++ub;
-                    if (++steps == 8) {
-                        uint64_t l2 = ub, h2 = whi;
-                        while (l2 < h2) {
-                            uint64_t m = (l2 + h2) >> 1;
-                            if (!join_less(mode, k, bk[m])) l2 = m + 1; else h2 = m;
-                        }
-                        ub = l2;
-                        break;
-                    }
-                }
-                counts[i] = (uint32_t)(ub - lb);
-                b_lo[i] = (uint32_t)lb;
-                lo = lb; /* equal next key re-uses the same lower bound */
+            uint64_t lb = lo;
+            /* upper bound */
+            hi = whi;
+            while (lo < hi) {
+                uint64_t m = (lo + hi) >> 1;
+                if (!join_less(mode, k, bk[m])) lo = m + 1; else hi = m;
             }
+            counts[i] = (uint32_t)(lo - lb);
+            b_lo[i] = (uint32_t)lb;
         }
     }
 }
